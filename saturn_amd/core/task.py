@@ -1,0 +1,236 @@
+"""Task and HParams: the user-facing job specification.
+
+Capability parity with the reference's ``saturn/core/representations/Task.py``
+(HParams: Task.py:23-62, Task: Task.py:65-179), redesigned for MI355X:
+
+- the task *spec* (model factory, dataloader factory, loss, hparams) is kept
+  separate from scheduler *bookkeeping* (``TaskProgress``), fixing the
+  reference's destructive mutation of ``strategies[g].runtime``
+  (reference executor.py:166-172);
+- ``epoch_length`` is computed lazily (the reference tokenizes the whole
+  corpus inside ``Task.__init__``, Task.py:127);
+- checkpoints carry optimizer state as well as model state (the reference
+  loses momentum/Adam moments at every interval boundary, DDP.py:181-182);
+- checkpoint files keep the reference's ``<save_dir>/<name>.pt`` naming
+  contract (Task.py:150-153).
+"""
+
+from __future__ import annotations
+
+import os
+import random
+import string
+import typing
+from typing import Any, Callable, Dict, List, Optional
+
+import torch
+
+
+class HParams:
+    """Training hyperparameters for a task.
+
+    Exactly one of ``epochs`` / ``batch_count`` must be given (reference
+    Task.py:42-44 enforces the same exclusivity).
+
+    Parameters
+    ----------
+    lr : learning rate.
+    epochs : number of dataloader passes (mutually exclusive with batch_count).
+    batch_count : total number of batches to train (mutually exclusive with
+        epochs).
+    optimizer_cls : optimizer class; instantiated by the executor.  When None
+        the executor defaults to the framework's fused SGD.
+    kwargs : forwarded to the user's ``get_model`` factory.
+    """
+
+    def __init__(
+        self,
+        lr: float,
+        epochs: Optional[int] = None,
+        batch_count: Optional[int] = None,
+        optimizer_cls: Optional[type] = None,
+        **kwargs: Any,
+    ) -> None:
+        if (batch_count is None) == (epochs is None):
+            raise ValueError(
+                "Exactly one of `epochs` and `batch_count` must be set."
+            )
+        self.lr = lr
+        self.epochs = epochs
+        self.batch_count = batch_count
+        self.optimizer_cls = optimizer_cls
+        self.kwargs = kwargs
+
+    def as_dict(self) -> Dict[str, Any]:
+        return {
+            "lr": self.lr,
+            "epochs": self.epochs,
+            "batch_count": self.batch_count,
+        }
+
+    def __repr__(self) -> str:
+        tail = (
+            f"epochs={self.epochs}"
+            if self.epochs is not None
+            else f"batch_count={self.batch_count}"
+        )
+        return f"HParams(lr={self.lr}, {tail})"
+
+
+def _random_name(k: int = 16) -> str:
+    return "".join(random.choices(string.ascii_uppercase + string.digits, k=k))
+
+
+class Task:
+    """A single training job submitted to the orchestrator.
+
+    Mirrors the reference ``Task`` contract (Task.py:65-179): callables for
+    model/dataloader construction (never pre-instantiated tensors), a loss
+    function, hyperparameters, an optional ``gpu_range`` to prune the search
+    space, free-form ``hints`` consumed by executors, and a ``save_dir`` for
+    the ``<name>.pt`` checkpoint.
+
+    Scheduler state lives in attributes the solver/engine manage:
+    ``strategies`` ({gpu_count -> Strategy}, filled by the trial runner),
+    ``selected_strategy`` (set by the solver), ``total_batches`` /
+    ``current_batch`` / ``batches_completed`` (advanced by the engine).
+    """
+
+    def __init__(
+        self,
+        get_model: Callable[..., torch.nn.Module],
+        get_dataloader: Callable[[], typing.Iterable],
+        loss_function: Callable,
+        hparams: HParams,
+        gpu_range: Optional[List[int]] = None,
+        name: Optional[str] = None,
+        hints: Optional[Dict[str, Any]] = None,
+        save_dir: str = "./saved_models",
+    ) -> None:
+        self.internal_get_model = get_model
+        self.internal_dl = get_dataloader
+        self.loss_function = loss_function
+        self.hparams = hparams
+        self.gpu_range = gpu_range
+        self.hints = hints or {}
+        self.name = name if name is not None else _random_name()
+        self.save_dir = save_dir
+        os.makedirs(save_dir, exist_ok=True)
+
+        if self.hints.get("is_transformer", False) and not isinstance(
+            self.hints.get("transformer_cls"), (set, frozenset)
+        ):
+            raise ValueError(
+                "A transformer task must pass its block class(es) as a set "
+                "in hints['transformer_cls'] (consumed by the FSDP wrap "
+                "policy)."
+            )
+
+        # Scheduler bookkeeping -------------------------------------------
+        self.strategies: Dict[int, "Strategy"] = {}  # noqa: F821
+        self.selected_strategy = None
+        self.current_batch = 0  # dataloader cursor within the epoch
+        self.batches_completed = 0  # scheduler bookkeeping (monotonic)
+        self._epoch_length: Optional[int] = None
+        self._total_batches: Optional[int] = (
+            hparams.batch_count if hparams.batch_count is not None else None
+        )
+
+    # -- dataloader bookkeeping -------------------------------------------
+    @property
+    def epoch_length(self) -> int:
+        """Batches per epoch.  Lazily computed (reference computes it
+        eagerly in ``__init__``, Task.py:127 — expensive and surprising)."""
+        if self._epoch_length is None:
+            self._epoch_length = len(self.internal_dl())
+        return self._epoch_length
+
+    @property
+    def total_batches(self) -> int:
+        if self._total_batches is None:
+            self._total_batches = self.epoch_length * self.hparams.epochs
+        return self._total_batches
+
+    @total_batches.setter
+    def total_batches(self, v: int) -> None:
+        self._total_batches = v
+
+    @property
+    def remaining_batches(self) -> int:
+        return max(0, self.total_batches - self.batches_completed)
+
+    def get_iterator(self, modified_dl=None):
+        """An iterator fast-forwarded to ``current_batch`` (reference
+        Task.py:132-140)."""
+        dl = iter(modified_dl) if modified_dl is not None else iter(self.internal_dl())
+        for _ in range(self.current_batch):
+            next(dl)
+        return dl
+
+    def get_fresh_iterator(self):
+        return iter(self.internal_dl())
+
+    def reconfigure(self, batch_count: int) -> None:
+        """Advance the dataloader cursor after an interval ran
+        ``batch_count`` batches (reference Task.py:155-157)."""
+        self.current_batch = (self.current_batch + batch_count) % self.epoch_length
+
+    # -- checkpointing -----------------------------------------------------
+    @property
+    def ckpt_path(self) -> str:
+        return os.path.join(self.save_dir, f"{self.name}.pt")
+
+    def has_ckpt(self) -> bool:
+        return os.path.isfile(self.ckpt_path)
+
+    def save_checkpoint(
+        self,
+        model: torch.nn.Module,
+        optimizer: Optional[torch.optim.Optimizer] = None,
+        extra: Optional[Dict[str, Any]] = None,
+    ) -> None:
+        """Checkpoint = model state + optimizer state (+ executor extras).
+
+        The reference saves only the model state dict and silently resets
+        optimizer moments at every interval boundary; we keep
+        both under the same ``<name>.pt`` path.
+        """
+        payload = {
+            "model": {k: v.cpu() for k, v in model.state_dict().items()},
+            "optimizer": optimizer.state_dict() if optimizer is not None else None,
+            "extra": extra or {},
+        }
+        tmp = self.ckpt_path + ".tmp"
+        torch.save(payload, tmp)
+        os.replace(tmp, self.ckpt_path)  # atomic: no torn ckpt on crash
+
+    def load_checkpoint(self) -> Optional[Dict[str, Any]]:
+        if not self.has_ckpt():
+            return None
+        return torch.load(self.ckpt_path, map_location="cpu", weights_only=False)
+
+    def get_model(self, fresh: bool = False) -> torch.nn.Module:
+        """Build the model; when a checkpoint exists (and not ``fresh``)
+        load its weights (reference Task.py:162-169)."""
+        if self.hparams.kwargs:
+            model = self.internal_get_model(self.hparams.kwargs)
+        else:
+            model = self.internal_get_model()
+        if not fresh:
+            ckpt = self.load_checkpoint()
+            if ckpt is not None:
+                model.load_state_dict(ckpt["model"])
+        return model
+
+    # -- strategy selection ------------------------------------------------
+    def select_strategy(self, strat) -> None:
+        self.selected_strategy = strat
+
+    def change_name(self, name: Optional[str] = None) -> None:
+        self.name = name if name is not None else _random_name()
+
+    def __repr__(self) -> str:
+        return (
+            f"Task({self.name}, {self.hparams!r}, "
+            f"selected={self.selected_strategy})"
+        )

@@ -1,0 +1,177 @@
+"""DDP executor: data-parallel training, one process per GPU over RCCL.
+
+Capability parity with the reference example UDP
+(``examples/wikitext103/executors/DDP.py:41-182``) with its selection bug
+fixed: ``search`` returns a real parameter dict on success (the reference
+returned ``(None, rt)`` and was therefore never selectable,
+DDP.py:71-72 vs PerformanceEvaluator.py:110).
+
+The gradient synchronization is this framework's own bucketed flat-buffer
+engine (``saturn_amd.parallel.ddp.BucketedDDP``) on RCCL-over-xGMI, and the
+optimizer step is the fused multi-tensor HIP kernel when running on GPU
+(``saturn_amd.ops``).
+"""
+
+from __future__ import annotations
+
+from timeit import default_timer as timer
+from typing import Any, Dict, List, Optional, Tuple
+
+from saturn_amd.core.technique import BaseTechnique
+from saturn_amd.executors.launch import (
+    destroy_process_group,
+    gang_spawn,
+    init_process_group,
+)
+
+TRIAL_BATCHES = 3  # 1 warmup + 2 timed (reference used 2: DDP.py:43)
+
+
+def _make_optimizer(task, model):
+    """Task-specified optimizer, defaulting to the framework's fused SGD on
+    GPU / torch SGD on CPU."""
+    import torch
+
+    params = list(model.parameters())
+    cls = task.hparams.optimizer_cls
+    if cls is None:
+        from saturn_amd.ops.optim import FusedSGD
+
+        return FusedSGD(params, lr=task.hparams.lr)
+    return cls(params, lr=task.hparams.lr)
+
+
+def _ddp_worker(
+    rank: int,
+    world: int,
+    task,
+    tid: int,
+    batch_count: int,
+    params: Optional[Dict[str, Any]],
+    trial: bool,
+):
+    """One rank of the DDP gang.  Returns (rank 0) the measured per-batch
+    seconds when ``trial`` else None."""
+    import torch
+
+    from saturn_amd.parallel.ddp import BucketedDDP
+
+    backend = init_process_group(rank, world)
+    try:
+        device = torch.device("cuda", rank) if backend == "nccl" else torch.device("cpu")
+        dtype = torch.bfloat16 if backend == "nccl" else torch.float32
+
+        model = task.get_model()
+        model = model.to(device=device, dtype=dtype)
+        model.train()
+        bucket_mb = float((params or {}).get("bucket_mb", 64.0))
+        ddp = BucketedDDP(model, bucket_mb=bucket_mb)
+        optimizer = _make_optimizer(task, model)
+
+        ckpt = task.load_checkpoint()
+        if ckpt is not None and ckpt.get("optimizer") is not None and not trial:
+            try:
+                optimizer.load_state_dict(ckpt["optimizer"])
+            except Exception:
+                pass  # optimizer class may have changed between intervals
+
+        it = task.get_iterator() if not trial else task.get_fresh_iterator()
+
+        def step(batch) -> None:
+            x, y = batch
+            x = x.to(device, non_blocking=True)
+            y = y.to(device, non_blocking=True)
+            out = ddp(x)
+            loss = task.loss_function(out, y)
+            loss.backward()
+            ddp.grad_sync()
+            optimizer.step()
+            ddp.zero_grad_buffers()
+
+        def next_batch():
+            nonlocal it
+            try:
+                return next(it)
+            except StopIteration:
+                it = task.get_fresh_iterator()
+                return next(it)
+
+        result = None
+        if trial:
+            n_timed = batch_count - 1
+            step(next_batch())  # warmup
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+            t0 = timer()
+            for _ in range(n_timed):
+                step(next_batch())
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+            result = (timer() - t0) / max(1, n_timed)
+        else:
+            for _ in range(batch_count):
+                step(next_batch())
+            if device.type == "cuda":
+                torch.cuda.synchronize()
+            if rank == 0:
+                task.save_checkpoint(model, optimizer)
+            import torch.distributed as dist
+
+            if world > 1:
+                dist.barrier()
+        return result
+    finally:
+        destroy_process_group()
+
+
+class DDPExecutor(BaseTechnique):
+    """Data parallelism via bucketed RCCL all-reduce."""
+
+    name = "ddp"
+
+    @staticmethod
+    def execute(task, gpus: List[int], tid: int, batch_count: int) -> None:
+        params = (
+            task.selected_strategy.parameters
+            if task.selected_strategy is not None
+            else None
+        )
+        gang_spawn(
+            _ddp_worker,
+            len(gpus),
+            tid,
+            task,
+            tid,
+            batch_count,
+            params,
+            False,
+        )
+
+    @staticmethod
+    def search(
+        task, gpus: List[int], tid: int
+    ) -> Tuple[Optional[Dict[str, Any]], float]:
+        """Time a short trial; tune the bucket size over a small grid.
+
+        Returns ({"bucket_mb": best}, seconds_per_batch) or (None, inf) when
+        the trial cannot run (e.g. OOM)."""
+        world = len(gpus)
+        candidates = [64.0] if world == 1 else [32.0, 128.0]
+        best: Tuple[Optional[Dict[str, Any]], float] = (None, float("inf"))
+        for mb in candidates:
+            try:
+                bt = gang_spawn(
+                    _ddp_worker,
+                    world,
+                    tid,
+                    task,
+                    tid,
+                    TRIAL_BATCHES,
+                    {"bucket_mb": mb},
+                    True,
+                )
+            except Exception:
+                continue
+            if bt is not None and bt < best[1]:
+                best = ({"bucket_mb": mb}, bt)
+        return best

@@ -1,0 +1,133 @@
+"""Bucketed data-parallel gradient synchronization for xGMI.
+
+MI355X-native replacement for ``torch.nn.parallel.DistributedDataParallel``
+as used by the reference's DDP executor (DDP.py:90,155): the hard systems
+work the reference delegated to torch's C++ reducer is done here explicitly,
+laid out for the MI355X node fabric:
+
+- gradients accumulate directly into one flat per-bucket buffer
+  (``param.grad`` is a view into it), so there is no pack/unpack kernel at
+  all — the bucket is RCCL-ready the moment its last grad lands;
+- buckets are built in reverse parameter order (backward completes roughly
+  in reverse) and all-reduced asynchronously as they fill, overlapping the
+  remaining backward;
+- the default bucket size targets the 7-link xGMI fabric: RCCL pipelines
+  multiple in-flight ring all-reduces across links, so several mid-size
+  buckets in flight beat one giant one.  Default 64 MiB.
+
+Runs on ``nccl`` (= RCCL) on GPU and on ``gloo`` for the CPU test suite —
+same code path, same hooks.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class _Bucket:
+    def __init__(self, params: List[torch.nn.Parameter], device, dtype) -> None:
+        self.params = params
+        numels = [p.numel() for p in params]
+        self.flat = torch.zeros(sum(numels), device=device, dtype=dtype)
+        # Carve grad views out of the flat buffer.
+        offset = 0
+        self.views = []
+        for p, n in zip(params, numels):
+            self.views.append(self.flat[offset : offset + n].view_as(p))
+            offset += n
+        self.pending = 0
+        self.work: Optional[dist.Work] = None
+
+    def attach_grads(self) -> None:
+        for p, v in zip(self.params, self.views):
+            p.grad = v
+
+    def reset(self) -> None:
+        self.pending = len(self.params)
+        self.work = None
+
+
+class BucketedDDP(torch.nn.Module):
+    """Wrap a module; forward passes through; gradients are averaged across
+    the process group during backward.  Call :meth:`grad_sync` after
+    ``loss.backward()`` and before ``optimizer.step()``."""
+
+    def __init__(
+        self,
+        module: torch.nn.Module,
+        bucket_mb: float = 64.0,
+        process_group=None,
+        grad_dtype: Optional[torch.dtype] = None,
+    ) -> None:
+        super().__init__()
+        self.module = module
+        self.pg = process_group
+        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+
+        # Broadcast initial parameters from rank 0 (reference relies on the
+        # DDP ctor for this, DDP.py:90).
+        if self.world > 1:
+            with torch.no_grad():
+                for p in module.parameters():
+                    dist.broadcast(p.data, src=0, group=self.pg)
+
+        params = [p for p in module.parameters() if p.requires_grad]
+        bucket_bytes = int(bucket_mb * 1024 * 1024)
+        self.buckets: List[_Bucket] = []
+        self._param_bucket = {}
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        for p in reversed(params):  # backward finishes roughly in reverse
+            if cur and (
+                cur_bytes + p.numel() * p.element_size() > bucket_bytes
+                or p.dtype != cur[0].dtype
+                or p.device != cur[0].device
+            ):
+                self._seal(cur, grad_dtype)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += p.numel() * p.element_size()
+        if cur:
+            self._seal(cur, grad_dtype)
+
+        for b in self.buckets:
+            b.attach_grads()
+            b.reset()
+            for p in b.params:
+                self._param_bucket[p] = b
+                p.register_post_accumulate_grad_hook(self._hook)
+
+    def _seal(self, params: List[torch.nn.Parameter], grad_dtype) -> None:
+        p0 = params[0]
+        self.buckets.append(
+            _Bucket(params, p0.device, grad_dtype or p0.dtype)
+        )
+
+    def _hook(self, p: torch.nn.Parameter) -> None:
+        b = self._param_bucket[p]
+        b.pending -= 1
+        if b.pending == 0 and self.world > 1:
+            b.work = dist.all_reduce(b.flat, async_op=True, group=self.pg)
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def grad_sync(self) -> None:
+        """Wait for in-flight bucket all-reduces and average."""
+        inv = 1.0 / self.world
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+            if self.world > 1:
+                b.flat.mul_(inv)
+            b.reset()
+
+    def zero_grad_buffers(self, set_to_none: bool = False) -> None:
+        """Zero the flat grad buffers (grads are views; never set to None)."""
+        for b in self.buckets:
+            b.flat.zero_()
+            b.attach_grads()
+            b.reset()

@@ -1,0 +1,175 @@
+"""Trial runner: empirically profile every (task, gpu-count, technique) cell.
+
+Parity with reference ``saturn/trial_runner/PerformanceEvaluator.py:21-115``:
+fan the cells out over the node's GPUs, call each technique's own
+``search()`` (which autotunes its parameters with short timed trials), and
+write the winning ``Strategy`` per (task, gpu-count) back onto the tasks.
+Differences from the reference:
+
+- cells are packed onto GPUs with a thread-pool + GPU-slot allocator instead
+  of Ray placement (``ray_search.options(num_gpus=g)``,
+  PerformanceEvaluator.py:74-84);
+- each cell runs in its own spawn subprocess with ``HIP_VISIBLE_DEVICES``
+  restricted to its gang, so an intentionally-OOMing trial cannot poison a
+  sibling (SURVEY §7 hard-part 5);
+- search returns per-batch seconds; whole-job runtime and the per-batch time
+  are both stored (``Strategy.batch_time``), feeding the solver's
+  remaining-work computation.
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+from timeit import default_timer as timer
+from typing import List, Optional
+
+from saturn_amd.core.strategy import INFEASIBLE_RUNTIME, Strategy
+from saturn_amd.engine.gang import call_in_subprocess
+from saturn_amd.library import retrieve
+from saturn_amd.solver.milp import detect_gpu_count
+
+log = logging.getLogger(__name__)
+
+
+class _GpuPool:
+    """Reserve ``g`` concrete GPU ids at a time (any ids — trials are
+    placement-agnostic on a fully-connected xGMI node)."""
+
+    def __init__(self, n: int) -> None:
+        self.free = set(range(n))
+        self.cv = threading.Condition()
+
+    def acquire(self, g: int) -> List[int]:
+        with self.cv:
+            while len(self.free) < g:
+                self.cv.wait()
+            ids = sorted(self.free)[:g]
+            self.free -= set(ids)
+            return ids
+
+    def release(self, ids: List[int]) -> None:
+        with self.cv:
+            self.free |= set(ids)
+            self.cv.notify_all()
+
+
+def _run_cell(executor, task, g: int, tid: int):
+    """Subprocess body for one trial cell."""
+    return executor.search(task, list(range(g)), tid)
+
+
+def search(
+    tasks: List,
+    executor_names: Optional[List[str]] = None,
+    log_level: bool = False,
+    n_gpus: Optional[int] = None,
+    trial_timeout: Optional[float] = 1800.0,
+    isolate: bool = True,
+) -> None:
+    """Profile all cells and attach per-gpu-count winning Strategies.
+
+    ``isolate=False`` runs cells in-process (CPU test mode — no HIP context
+    to isolate)."""
+    if log_level:
+        logging.basicConfig(
+            format="%(asctime)s %(levelname)-8s %(message)s",
+            level=logging.INFO,
+            datefmt="%Y-%m-%d %H:%M:%S",
+        )
+    executors = retrieve(executor_names)
+    if n_gpus is None:
+        n_gpus = detect_gpu_count()
+    default_range = list(range(1, n_gpus + 1))
+
+    cells = []  # (task_idx, g, exec_idx)
+    for ti, t in enumerate(tasks):
+        for g in t.gpu_range or default_range:
+            if g > n_gpus:
+                continue
+            for ei in range(len(executors)):
+                cells.append((ti, g, ei))
+
+    results = {}
+    pool = _GpuPool(n_gpus)
+    lock = threading.Lock()
+
+    def run(cell_idx: int) -> None:
+        ti, g, ei = cells[cell_idx]
+        task, executor = tasks[ti], executors[ei]
+        ids = pool.acquire(g)
+        t0 = timer()
+        try:
+            if isolate:
+                visible = ",".join(str(i) for i in ids)
+                params, bt = call_in_subprocess(
+                    _run_cell,
+                    executor,
+                    task,
+                    g,
+                    cell_idx,
+                    env={
+                        "HIP_VISIBLE_DEVICES": visible,
+                        "CUDA_VISIBLE_DEVICES": visible,
+                    },
+                    timeout=trial_timeout,
+                )
+            else:
+                params, bt = executor.search(task, list(range(g)), cell_idx)
+        except Exception as e:  # infeasible / crashed cell
+            log.info(
+                "trial (%s, %dG, %s) failed: %s",
+                task.name,
+                g,
+                getattr(executor, "name", executor),
+                e,
+            )
+            params, bt = None, float("inf")
+        finally:
+            pool.release(ids)
+        with lock:
+            results[(ti, g, ei)] = (params, bt)
+        log.info(
+            "trial (%s, %dG, %s): params=%s batch_time=%.4fs (%.1fs trial)",
+            task.name,
+            g,
+            getattr(executor, "name", executor),
+            params,
+            bt if bt == bt else -1.0,
+            timer() - t0,
+        )
+
+    log.info("%d trial cells to run on %d GPUs", len(cells), n_gpus)
+    threads = [
+        threading.Thread(target=run, args=(i,), daemon=True)
+        for i in range(len(cells))
+    ]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+
+    # Initialize every cell with the infeasible sentinel, then pick winners
+    # (reference PerformanceEvaluator.py:96-115).
+    for t in tasks:
+        for g in default_range:
+            t.strategies[g] = Strategy(None, g, None, INFEASIBLE_RUNTIME)
+    for ti, t in enumerate(tasks):
+        for g in t.gpu_range or default_range:
+            if g > n_gpus:
+                continue
+            best = None  # (bt, executor, params)
+            for ei, ex in enumerate(executors):
+                params, bt = results.get((ti, g, ei), (None, float("inf")))
+                if params is not None and bt == bt and bt != float("inf"):
+                    if best is None or bt < best[0]:
+                        best = (bt, ex, params)
+            if best is not None:
+                bt, ex, params = best
+                t.strategies[g] = Strategy(
+                    ex,
+                    g,
+                    params,
+                    runtime=bt * t.total_batches,
+                    batch_time=bt,
+                )

@@ -1,0 +1,134 @@
+"""Engine + forecast + end-to-end orchestration on CPU (gloo).
+
+BASELINE.json config 1: "2-job MLP lr-sweep on CPU/gloo" — the plumbing test
+the reference never had.  Exercises: subprocess gangs,
+torch.distributed over gloo with world_size 2, the bucketed DDP engine,
+dependency-ordered interval execution, checkpoint/resume across intervals,
+and the full orchestrate() loop.
+"""
+
+import os
+
+import pytest
+import torch
+
+from saturn_amd import HParams, Strategy, Task, orchestrate, register, search
+from saturn_amd.engine import call_in_subprocess, forecast
+from saturn_amd.executors.ddp import DDPExecutor, _ddp_worker
+from saturn_amd.executors.launch import gang_spawn
+from saturn_amd.models import get_mlp_dataloader, get_mlp_model, mse_loss
+from saturn_amd.solver import solve
+
+
+def make_mlp_task(name, save_dir, lr=1e-2, batch_count=6, gpu_range=None):
+    return Task(
+        get_mlp_model,
+        get_mlp_dataloader,
+        mse_loss,
+        HParams(lr=lr, batch_count=batch_count),
+        gpu_range=gpu_range or [1, 2],
+        name=name,
+        save_dir=save_dir,
+    )
+
+
+def test_call_in_subprocess_roundtrip():
+    assert call_in_subprocess(sorted, [3, 1, 2], timeout=120) == [1, 2, 3]
+
+
+def test_call_in_subprocess_error_propagates():
+    def boom():
+        raise ValueError("inner detail 42")
+
+    with pytest.raises(RuntimeError, match="inner detail 42"):
+        call_in_subprocess(boom, timeout=120)
+
+
+def test_gang_spawn_world2():
+    def f(rank, world, x):
+        return (rank, world, x) if rank == 0 else None
+
+    assert gang_spawn(f, 2, 900, 7, timeout=120) == (0, 2, 7)
+
+
+def test_ddp_worker_world2_gloo(save_dir):
+    """Two gloo ranks train the MLP; rank 0 checkpoints; weights identical
+    to a single-process run with the same effective data order is not
+    required — just finite loss + ckpt exists."""
+    t = make_mlp_task("g2", save_dir, batch_count=4)
+    gang_spawn(_ddp_worker, 2, 901, t, 901, 4, {"bucket_mb": 1.0}, False, timeout=300)
+    assert t.has_ckpt()
+    ck = t.load_checkpoint()
+    assert ck["optimizer"] is not None
+
+
+def test_ddp_search_returns_params(save_dir, library_path):
+    t = make_mlp_task("s1", save_dir)
+    params, bt = DDPExecutor.search(t, [0], 902)
+    assert params is not None and "bucket_mb" in params
+    assert 0 < bt < 60
+
+
+def test_trial_runner_search_fills_strategies(save_dir, library_path):
+    register("ddp", DDPExecutor)
+    a = make_mlp_task("tr_a", save_dir)
+    b = make_mlp_task("tr_b", save_dir, lr=1e-3)
+    search([a, b], log_level=False, n_gpus=2, isolate=False)
+    for t in (a, b):
+        for g in (1, 2):
+            assert g in t.strategies
+            s = t.strategies[g]
+            assert s.feasible, f"cell ({t.name},{g}) should be feasible"
+            assert s.batch_time > 0
+
+
+def test_forecast_quota_and_completion(save_dir):
+    t = make_mlp_task("f1", save_dir, batch_count=10)
+    t.strategies[1] = Strategy(DDPExecutor, 1, {"bucket_mb": 1}, 10.0, batch_time=1.0)
+    plan = solve([t], n_gpus=1, timeout=5)
+    t.select_strategy(t.strategies[1])
+    relevant, batches, completing = forecast([t], 5.0, plan)
+    assert relevant == [t]
+    assert batches == [5]
+    assert not completing
+    assert t.batches_completed == 5
+    relevant, batches, completing = forecast([t], 50.0, plan)
+    assert batches == [5]
+    assert completing == {t}
+
+
+def test_forecast_slow_task_still_progresses(save_dir):
+    # per-batch time longer than the interval: must still run >= 1 batch
+    t = make_mlp_task("f2", save_dir, batch_count=3)
+    t.strategies[1] = Strategy(DDPExecutor, 1, {"bucket_mb": 1}, 300.0, batch_time=100.0)
+    plan = solve([t], n_gpus=1, timeout=5)
+    t.select_strategy(t.strategies[1])
+    relevant, batches, _ = forecast([t], 10.0, plan)
+    assert batches == [1]
+
+
+def test_orchestrate_end_to_end_cpu(save_dir, library_path):
+    """The full loop: register -> search -> orchestrate, 2-job MLP lr sweep
+    on a fake 2-GPU node over gloo.  Models must checkpoint and every task
+    must complete its batch quota."""
+    register("ddp", DDPExecutor)
+    tasks = [
+        make_mlp_task("sweep_lr_a", save_dir, lr=1e-2, batch_count=6),
+        make_mlp_task("sweep_lr_b", save_dir, lr=1e-3, batch_count=6),
+    ]
+    search(tasks, n_gpus=2, isolate=False)
+    # short intervals so multiple solve/execute cycles happen
+    max_bt = max(s.batch_time for t in tasks for s in t.strategies.values() if s.batch_time)
+    orchestrate(tasks, interval=max_bt * 4, n_gpus=2, solver_timeout=5)
+    for t in tasks:
+        assert t.remaining_batches == 0
+        assert t.has_ckpt()
+
+
+def test_orchestrate_respects_gpu_range(save_dir, library_path):
+    register("ddp", DDPExecutor)
+    t = make_mlp_task("one_gpu", save_dir, batch_count=3, gpu_range=[1])
+    search([t], n_gpus=2, isolate=False)
+    assert t.strategies[1].feasible
+    orchestrate([t], interval=1e6, n_gpus=2, solver_timeout=5)
+    assert t.remaining_batches == 0

@@ -1,0 +1,149 @@
+"""Solver unit tests on synthetic runtime tables (no GPUs, no models).
+
+The solver consumes only (gpu_count, runtime) tuples (reference
+milp.py:70-81), so plan validity is checkable end-to-end on fakes:
+one strategy per task, occupancy matches the chosen gpu count, no temporal
+overlap on shared GPUs, makespan >= every task's end.
+"""
+
+import pytest
+
+from saturn_amd.core import Strategy, Task, HParams
+from saturn_amd.solver import Plan, apply_plan, solve
+from saturn_amd.solver.milp import _greedy_plan
+
+
+class FakeExec:
+    name = "fake"
+
+
+def make_task(name, runtimes, save_dir, batch_count=100):
+    """runtimes: {gpu_count: whole-job seconds}."""
+    t = Task(
+        get_model=lambda: None,
+        get_dataloader=lambda: [],
+        loss_function=lambda o, y: None,
+        hparams=HParams(lr=1e-3, batch_count=batch_count),
+        name=name,
+        save_dir=save_dir,
+    )
+    for g, r in runtimes.items():
+        t.strategies[g] = Strategy(
+            FakeExec, g, {"p": 1}, runtime=r, batch_time=r / batch_count
+        )
+    return t
+
+
+def check_plan_valid(plan: Plan, task_list, n_gpus):
+    assert plan.task_names == [t.name for t in task_list]
+    for i, t in enumerate(task_list):
+        g = plan.gpu_counts[i]
+        assert g in t.strategies
+        assert len(plan.gpu_sets[i]) == g
+        assert all(0 <= x < n_gpus for x in plan.gpu_sets[i])
+        end = plan.start_times[i] + plan.runtimes[i]
+        assert plan.makespan >= end - 1e-6
+    # no overlap on shared GPUs
+    n = len(task_list)
+    for i in range(n):
+        for j in range(i + 1, n):
+            if set(plan.gpu_sets[i]) & set(plan.gpu_sets[j]):
+                si, ei = plan.start_times[i], plan.start_times[i] + plan.runtimes[i]
+                sj, ej = plan.start_times[j], plan.start_times[j] + plan.runtimes[j]
+                assert ei <= sj + 1e-6 or ej <= si + 1e-6, (
+                    f"tasks {i},{j} overlap on shared GPUs"
+                )
+
+
+def test_single_task_picks_fastest(save_dir):
+    t = make_task("a", {1: 100.0, 2: 60.0, 4: 40.0}, save_dir)
+    plan = solve([t], n_gpus=4, timeout=10)
+    check_plan_valid(plan, [t], 4)
+    assert plan.gpu_counts[0] == 4
+    assert plan.makespan == pytest.approx(40.0, rel=1e-3)
+
+
+def test_two_tasks_pack_in_parallel(save_dir):
+    # Two tasks, each 2-GPU/50s on a 4-GPU node: optimal packs them side by
+    # side (makespan 50), not sequentially (100).
+    a = make_task("a", {2: 50.0, 4: 40.0}, save_dir)
+    b = make_task("b", {2: 50.0, 4: 40.0}, save_dir)
+    plan = solve([a, b], n_gpus=4, timeout=20)
+    check_plan_valid(plan, [a, b], 4)
+    assert plan.makespan == pytest.approx(50.0, rel=1e-3)
+
+
+def test_sequential_when_node_too_small(save_dir):
+    a = make_task("a", {4: 30.0}, save_dir)
+    b = make_task("b", {4: 20.0}, save_dir)
+    plan = solve([a, b], n_gpus=4, timeout=20)
+    check_plan_valid(plan, [a, b], 4)
+    assert plan.makespan == pytest.approx(50.0, rel=1e-3)
+    deps = plan.dependency_dict()
+    # one of them must depend on the other
+    assert deps[0] or deps[1]
+
+
+def test_mixed_batch(save_dir):
+    tasks = [
+        make_task("a", {1: 80.0, 2: 45.0, 4: 30.0}, save_dir),
+        make_task("b", {1: 80.0, 2: 45.0, 4: 30.0}, save_dir),
+        make_task("c", {1: 20.0, 2: 12.0}, save_dir),
+        make_task("d", {1: 20.0, 2: 12.0}, save_dir),
+    ]
+    plan = solve(tasks, n_gpus=4, timeout=30)
+    check_plan_valid(plan, tasks, 4)
+    # sanity: beats the trivially sequential-on-best-option schedule
+    seq = sum(min(r for r in t.strategies and [s.runtime for s in t.strategies.values()]) for t in tasks)
+    assert plan.makespan <= seq + 1e-6
+
+
+def test_hysteresis_keeps_old_plan(save_dir):
+    t = make_task("a", {1: 5000.0}, save_dir)
+    old = solve([t], n_gpus=1, timeout=10)
+    old_ms = old.makespan
+    # re-solve with presolved: identical cost -> keep (shifted) old plan
+    plan2 = solve([t], presolved=old, interval=1000, n_gpus=1, timeout=10)
+    assert plan2 is old
+    assert plan2.makespan == pytest.approx(old_ms - 1000.0)
+
+
+def test_hysteresis_adopts_much_better_plan(save_dir):
+    t = make_task("a", {1: 10000.0}, save_dir)
+    old = solve([t], n_gpus=1, timeout=10)
+    # task suddenly much cheaper (e.g. most batches done)
+    t.batches_completed = 90
+    plan2 = solve([t], presolved=old, interval=1000, n_gpus=1, timeout=10)
+    assert plan2 is not old
+    assert plan2.makespan < 2000.0
+
+
+def test_infeasible_cells_avoided(save_dir):
+    t = make_task("a", {2: 50.0}, save_dir)
+    # add an infeasible 1-GPU cell (sentinel runtime, no params)
+    t.strategies[1] = Strategy(None, 1, None, 1e6)
+    plan = solve([t], n_gpus=2, timeout=10)
+    assert plan.gpu_counts[0] == 2
+
+
+def test_greedy_fallback_valid(save_dir):
+    tasks = [
+        make_task(f"t{i}", {1: 10.0 + i, 2: 6.0 + i, 4: 4.0 + i}, save_dir)
+        for i in range(6)
+    ]
+    plan = _greedy_plan(tasks, 4)
+    check_plan_valid(plan, tasks, 4)
+
+
+def test_apply_plan_selects_strategy(save_dir):
+    t = make_task("a", {1: 100.0, 2: 60.0}, save_dir)
+    plan = solve([t], n_gpus=2, timeout=10)
+    apply_plan([t], plan)
+    assert t.selected_strategy is t.strategies[plan.gpu_counts[0]]
+
+
+def test_remaining_work_shrinks_runtime(save_dir):
+    t = make_task("a", {1: 100.0}, save_dir, batch_count=100)
+    t.batches_completed = 50
+    plan = solve([t], n_gpus=1, timeout=10)
+    assert plan.makespan == pytest.approx(50.0, rel=1e-3)
