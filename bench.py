@@ -1,0 +1,149 @@
+"""Flagship benchmark: GPT-J-6B bf16 data-parallel training step.
+
+This measures the reference's headline workload (GPT-J-6B
+fine-tune, ctx 512, batch 8 per GPU, SGD — simple-verification.py:58-73) on
+saturn_amd's own training stack: fused CDNA4 kernels for
+LayerNorm/RoPE/attention/CE/optimizer, bucketed flat-buffer DDP over RCCL.
+
+Contract (driver):
+    python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run; ranks read
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.  Rank 0 prints ONE JSON
+line; `value` is whole-job samples/sec aggregated over all N GPUs;
+per-GPU work is fixed (weak scaling).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+from timeit import default_timer as timer
+
+import torch
+
+
+def log(*a):
+    print(*a, file=sys.stderr, flush=True)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--batch", type=int, default=8, help="per-GPU batch size")
+    ap.add_argument("--seq", type=int, default=512)
+    ap.add_argument("--layers", type=int, default=28)
+    ap.add_argument("--model", type=str, default="gptj-6b")
+    ap.add_argument("--bucket-mb", type=float, default=64.0)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    use_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", local_rank) if use_gpu else torch.device("cpu")
+    dtype = torch.bfloat16 if use_gpu else torch.float32
+
+    import torch.distributed as dist
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group(
+            backend="nccl" if use_gpu else "gloo", rank=rank, world_size=world
+        )
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        # the bench must run on our kernels, never a silent fallback
+        from saturn_amd.ops import require_ext
+
+        require_ext()
+
+    from saturn_amd.models.gptj import GPTJConfig, GPTJForCausalLM, pretraining_loss
+    from saturn_amd.ops.optim import FusedSGD
+    from saturn_amd.parallel.ddp import BucketedDDP
+
+    cfg = GPTJConfig(n_layer=args.layers, n_ctx=args.seq)
+    torch.manual_seed(1234)
+    log(f"[rank {rank}] building GPT-J ({cfg.n_layer} layers) on {device}...")
+    model = GPTJForCausalLM(cfg).to(device=device, dtype=dtype)
+    model.train()
+    n_params = sum(p.numel() for p in model.parameters())
+    ddp = BucketedDDP(model, bucket_mb=args.bucket_mb)
+    opt = FusedSGD(model.parameters(), lr=1e-5)
+
+    g = torch.Generator(device="cpu").manual_seed(4321 + rank)
+    x = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), generator=g).to(device)
+
+    def step() -> None:
+        logits = ddp(x)
+        loss = pretraining_loss(logits, x)
+        loss.backward()
+        ddp.grad_sync()
+        opt.step()
+        ddp.zero_grad_buffers()
+
+    log(f"[rank {rank}] {n_params/1e9:.2f}B params; warmup {args.warmup} steps")
+    for _ in range(args.warmup):
+        step()
+
+    if world > 1:
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = timer()
+    for _ in range(args.steps):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t1 = timer()
+
+    elapsed = t1 - t0
+    # MAX over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        samples_per_s = args.batch * world * args.steps / elapsed
+        print(
+            json.dumps(
+                {
+                    "metric": "samples_per_s",
+                    "value": samples_per_s,
+                    "unit": "samples/s",
+                    "n_gpus": world,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": ms_per_step,
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "bf16" if use_gpu else "fp32",
+                    "data": "synthetic",
+                    "config": {
+                        "model": args.model,
+                        "n_params_b": round(n_params / 1e9, 2),
+                        "global_batch": args.batch * world,
+                        "seq_len": args.seq,
+                        "parallelism": f"dp{world}",
+                        "optimizer": "fused_sgd",
+                    },
+                }
+            ),
+            flush=True,
+        )
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,3 +1,22 @@
 from .mlp import MLP, get_mlp_dataloader, get_mlp_model, mse_loss
+from .gpt2 import GPT2ForCausalLM, get_gpt2_model, gpt2_loss
+from .gptj import (
+    GPTJForCausalLM,
+    get_gptj_model,
+    make_token_dataloader,
+    pretraining_loss,
+)
 
-__all__ = ["MLP", "get_mlp_model", "get_mlp_dataloader", "mse_loss"]
+__all__ = [
+    "MLP",
+    "get_mlp_model",
+    "get_mlp_dataloader",
+    "mse_loss",
+    "GPT2ForCausalLM",
+    "get_gpt2_model",
+    "gpt2_loss",
+    "GPTJForCausalLM",
+    "get_gptj_model",
+    "make_token_dataloader",
+    "pretraining_loss",
+]

@@ -1,0 +1,37 @@
+// saturn_amd._C: pybind entry for the hand-written CDNA4 kernels.
+
+#include <torch/extension.h>
+#include <vector>
+
+namespace samd {
+void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+               std::vector<at::Tensor> moms, double lr, double momentum,
+               double weight_decay);
+void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, double bc1, double bc2);
+std::vector<at::Tensor> norm_fwd(at::Tensor x, at::Tensor w,
+                                 c10::optional<at::Tensor> b, double eps,
+                                 bool rms);
+std::vector<at::Tensor> norm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                 at::Tensor mean, at::Tensor rstd, bool rms,
+                                 bool needs_db);
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
+                               int64_t Tr, int64_t ignore_index);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                  at::Tensor dloss, int64_t Tr, int64_t ignore_index);
+void rope_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
+                bool half_style, bool backward);
+}  // namespace samd
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "saturn_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("fused_sgd", &samd::fused_sgd, "fused multi-tensor SGD");
+  m.def("fused_adam", &samd::fused_adam, "fused multi-tensor AdamW");
+  m.def("norm_fwd", &samd::norm_fwd, "LayerNorm/RMSNorm forward");
+  m.def("norm_bwd", &samd::norm_bwd, "LayerNorm/RMSNorm backward");
+  m.def("ce_fwd", &samd::ce_fwd, "fused cross-entropy forward");
+  m.def("ce_bwd", &samd::ce_bwd, "fused cross-entropy backward");
+  m.def("rope_apply", &samd::rope_apply, "fused rotary embedding (in-place)");
+}

@@ -1,0 +1,63 @@
+"""Flash attention dispatch (K2).
+
+``saturn_amd._C.attn_fwd/attn_bwd`` are the hand-written CDNA4 MFMA kernels
+(attention.hip).  Until they are built into the extension, the GPU path
+falls back to the explicit-GEMM math composition (rocBLAS GEMMs + softmax —
+no Triton, no aotriton) and WARNS once: measurements taken on the fallback
+are not this framework's attention numbers.
+"""
+
+from __future__ import annotations
+
+import logging
+
+import torch
+
+from saturn_amd.ops import require_ext
+
+log = logging.getLogger(__name__)
+_warned = False
+
+
+class _FlashFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        ext = require_ext()
+        o, lse = ext.attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        ext = require_ext()
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = ext.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.causal)
+        return dq, dk, dv, None
+
+
+def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
+    """q,k,v: [B, H, T, D] bf16/fp16 contiguous (GQA: H_kv may divide H)."""
+    ext = require_ext()
+    if hasattr(ext, "attn_fwd"):
+        if k.shape[1] != q.shape[1]:  # GQA: expand kv heads
+            rep = q.shape[1] // k.shape[1]
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        return _FlashFn.apply(
+            q.contiguous(), k.contiguous(), v.contiguous(), causal
+        )
+    global _warned
+    if not _warned:
+        log.warning(
+            "attn_fwd not in saturn_amd._C — using explicit-GEMM math "
+            "attention (slower; rebuild with attention.hip)"
+        )
+        _warned = True
+    from saturn_amd.ops.functional import attention_math
+
+    if k.shape[1] != q.shape[1]:
+        rep = q.shape[1] // k.shape[1]
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    return attention_math(q, k, v, causal=causal)

@@ -1,0 +1,243 @@
+"""Autograd wrappers over the CDNA4 kernels, with CPU reference fallbacks.
+
+GPU tensors route to ``saturn_amd._C`` (and raise loudly if the extension is
+missing — ops/__init__.require_ext); CPU tensors use the equivalent pure
+PyTorch math so the orchestration suite runs in the no-GPU container.  The
+numerics tests compare the two paths (tests/test_ops_gpu.py).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from saturn_amd.ops import require_ext
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm / RMSNorm (K4)
+# ---------------------------------------------------------------------------
+class _NormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, eps, rms):
+        ext = require_ext()
+        x = x.contiguous()
+        y, mean, rstd = ext.norm_fwd(x, w, b, eps, rms)
+        ctx.save_for_backward(x, w, mean, rstd)
+        ctx.rms = rms
+        ctx.has_b = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        x, w, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ext.norm_bwd(
+            dy.contiguous(), x, w, mean, rstd, ctx.rms, ctx.has_b
+        )
+        return (
+            dx,
+            dw.to(w.dtype),
+            db.to(w.dtype) if ctx.has_b else None,
+            None,
+            None,
+        )
+
+
+def fused_layer_norm(x, weight, bias=None, eps: float = 1e-5):
+    if x.is_cuda:
+        return _NormFn.apply(x, weight, bias, eps, False)
+    return F.layer_norm(
+        x.float(), (x.shape[-1],),
+        weight.float(), bias.float() if bias is not None else None, eps
+    ).to(x.dtype)
+
+
+def fused_rms_norm(x, weight, eps: float = 1e-6):
+    if x.is_cuda:
+        return _NormFn.apply(x, weight, None, eps, True)
+    xf = x.float()
+    y = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (y * weight.float()).to(x.dtype)
+
+
+class FusedLayerNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5, bias: bool = True):
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(torch.ones(dim))
+        self.bias = torch.nn.Parameter(torch.zeros(dim)) if bias else None
+
+    def forward(self, x):
+        return fused_layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class FusedRMSNorm(torch.nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-6):
+        super().__init__()
+        self.eps = eps
+        self.weight = torch.nn.Parameter(torch.ones(dim))
+
+    def forward(self, x):
+        return fused_rms_norm(x, self.weight, self.eps)
+
+
+# ---------------------------------------------------------------------------
+# Fused shift + cross-entropy (K8)
+# ---------------------------------------------------------------------------
+class _CeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets_flat, Tr, ignore_index):
+        ext = require_ext()
+        loss, lse = ext.ce_fwd(logits, targets_flat, Tr, ignore_index)
+        valid = (targets_flat != ignore_index).sum().clamp(min=1)
+        ctx.save_for_backward(logits, targets_flat, lse, valid)
+        ctx.Tr = Tr
+        ctx.ignore_index = ignore_index
+        return loss.sum() / valid.float()
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        ext = require_ext()
+        logits, targets_flat, lse, valid = ctx.saved_tensors
+        dloss = torch.full_like(lse, 0.0)
+        dloss.fill_(1.0)
+        dloss = dloss * (grad_out.float() / valid.float())
+        dlogits = ext.ce_bwd(
+            logits, targets_flat, lse, dloss, ctx.Tr, ctx.ignore_index
+        )
+        return dlogits, None, None, None
+
+
+def fused_cross_entropy(
+    logits: torch.Tensor,
+    targets: torch.Tensor,
+    shift: bool = True,
+    ignore_index: int = -100,
+) -> torch.Tensor:
+    """Causal-LM loss: mean CE of ``logits[:, :-1]`` vs ``targets[:, 1:]``
+    (``shift=True``) without materializing shifted copies of the logits.
+
+    logits: [B, T, V]; targets: [B, T] int64.
+    """
+    B, T, V = logits.shape
+    if logits.is_cuda:
+        Tr = T - 1 if shift else T
+        tg = targets[:, 1:] if shift else targets
+        return _CeFn.apply(
+            logits.contiguous(), tg.contiguous().view(-1), Tr, ignore_index
+        )
+    # CPU reference
+    if shift:
+        lg = logits[:, :-1].float().reshape(-1, V)
+        tg = targets[:, 1:].reshape(-1)
+    else:
+        lg = logits.float().reshape(-1, V)
+        tg = targets.reshape(-1)
+    return F.cross_entropy(lg, tg, ignore_index=ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# Rotary embedding (K3)
+# ---------------------------------------------------------------------------
+def rope_tables(
+    T: int,
+    rotary_dim: int,
+    base: float = 10000.0,
+    device=None,
+    interleaved: bool = True,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """cos/sin tables [T, rotary_dim/2] fp32."""
+    half = rotary_dim // 2
+    inv = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float32) * 2 / rotary_dim))
+    t = torch.arange(T, dtype=torch.float32)
+    freqs = torch.outer(t, inv)
+    return freqs.cos().to(device), freqs.sin().to(device)
+
+
+class _RopeFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos_bt, sin_bt, half_style):
+        ext = require_ext()
+        y = x.contiguous().clone()
+        ext.rope_apply(y, cos_bt, sin_bt, half_style, False)
+        ctx.save_for_backward(cos_bt, sin_bt)
+        ctx.half_style = half_style
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        cos_bt, sin_bt = ctx.saved_tensors
+        dx = dy.contiguous().clone()
+        ext.rope_apply(dx, cos_bt, sin_bt, ctx.half_style, True)
+        return dx, None, None, None
+
+
+def _rope_cpu(x, cos_bt, sin_bt, half_style):
+    B, T, H, D = x.shape
+    half = cos_bt.shape[-1]
+    c = cos_bt.view(B, T, 1, half).float()
+    s = sin_bt.view(B, T, 1, half).float()
+    xf = x.float()
+    out = xf.clone()
+    if half_style:
+        x0 = xf[..., :half]
+        x1 = xf[..., half : 2 * half]
+        out[..., :half] = x0 * c - x1 * s
+        out[..., half : 2 * half] = x1 * c + x0 * s
+    else:
+        x0 = xf[..., 0 : 2 * half : 2]
+        x1 = xf[..., 1 : 2 * half : 2]
+        out[..., 0 : 2 * half : 2] = x0 * c - x1 * s
+        out[..., 1 : 2 * half : 2] = x1 * c + x0 * s
+    return out.to(x.dtype)
+
+
+def apply_rope(
+    x: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    half_style: bool = False,
+) -> torch.Tensor:
+    """x: [B, T, H, D]; cos/sin: [T, half] (broadcast over batch).
+
+    GPT-J uses interleaved pairs (half_style=False, GPTJ.py:56-61);
+    Llama/NeoX uses the half-split layout (half_style=True).
+    """
+    B, T, H, D = x.shape
+    half = cos.shape[-1]
+    cos_bt = cos.unsqueeze(0).expand(B, T, half).contiguous()
+    sin_bt = sin.unsqueeze(0).expand(B, T, half).contiguous()
+    if x.is_cuda:
+        return _RopeFn.apply(x, cos_bt, sin_bt, half_style)
+    return _rope_cpu(x, cos_bt, sin_bt, half_style)
+
+
+# ---------------------------------------------------------------------------
+# Causal attention (K2) — flash kernel lands in attention.hip; the math
+# fallback below (explicit GEMMs + softmax, fp32 accum like the reference's
+# GPTJ.py:164-191) is used on CPU and as the numerics reference.
+# ---------------------------------------------------------------------------
+def attention_math(q, k, v, causal: bool = True) -> torch.Tensor:
+    """q,k,v: [B, H, T, D].  fp32 score math, returns q.dtype."""
+    scale = 1.0 / (q.shape[-1] ** 0.5)
+    s = torch.matmul(q.float(), k.float().transpose(-1, -2)) * scale
+    if causal:
+        T = q.shape[-2]
+        mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.matmul(p, v.float()).to(q.dtype)
+
+
+def causal_attention(q, k, v) -> torch.Tensor:
+    """Dispatch: fused CDNA4 flash kernel on GPU (when built), math path on
+    CPU."""
+    if q.is_cuda:
+        from saturn_amd.ops import flash  # local import: optional kernel
+
+        return flash.flash_attention(q, k, v, causal=True)
+    return attention_math(q, k, v, causal=True)

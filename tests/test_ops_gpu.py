@@ -1,0 +1,229 @@
+"""Numerics tests for the hand-written CDNA4 kernels vs plain PyTorch fp32
+references (the contract: every HIP kernel is compared against an fp32
+reference of the same op)."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def requires_ext():
+    from saturn_amd.ops import require_ext
+
+    return require_ext()
+
+
+def rel_err(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).norm() / b.norm().clamp(min=1e-12)).item()
+
+
+# ---------------------------------------------------------------------------
+# Fused optimizers (K9)
+# ---------------------------------------------------------------------------
+def test_fused_sgd_matches_reference():
+    requires_ext()
+    from saturn_amd.ops.optim import FusedSGD
+
+    torch.manual_seed(0)
+    shapes = [(128, 64), (1000,), (33, 7), (4096,)]
+    ps_f32 = [torch.randn(s, device="cuda") for s in shapes]
+    gs = [torch.randn(s, device="cuda") for s in shapes]
+    ps = [p.clone().requires_grad_(True) for p in ps_f32]
+    for p, gr in zip(ps, gs):
+        p.grad = gr.clone()
+    opt = FusedSGD(ps, lr=0.1, momentum=0.9, weight_decay=0.01)
+    ref = torch.optim.SGD(
+        [p.clone().requires_grad_(True) for p in ps_f32],
+        lr=0.1, momentum=0.9, weight_decay=0.01,
+    )
+    for rp, gr in zip(ref.param_groups[0]["params"], gs):
+        rp.grad = gr.clone()
+    for _ in range(3):
+        opt.step()
+        ref.step()
+    for p, rp in zip(ps, ref.param_groups[0]["params"]):
+        assert rel_err(p, rp) < 1e-5
+
+
+def test_fused_sgd_bf16():
+    requires_ext()
+    from saturn_amd.ops.optim import FusedSGD
+
+    torch.manual_seed(0)
+    p32 = torch.randn(5000, device="cuda")
+    g32 = torch.randn(5000, device="cuda")
+    p = p32.to(torch.bfloat16).requires_grad_(True)
+    p.grad = g32.to(torch.bfloat16)
+    FusedSGD([p], lr=0.5).step()
+    ref = (p32.to(torch.bfloat16).float() - 0.5 * g32.to(torch.bfloat16).float())
+    assert rel_err(p, ref.to(torch.bfloat16)) < 1e-2
+
+
+def test_fused_adam_matches_adamw():
+    requires_ext()
+    from saturn_amd.ops.optim import FusedAdam
+
+    torch.manual_seed(0)
+    shapes = [(64, 32), (777,)]
+    init = [torch.randn(s, device="cuda") for s in shapes]
+    gs = [torch.randn(s, device="cuda") for s in shapes]
+    ps = [p.clone().requires_grad_(True) for p in init]
+    rps = [p.clone().requires_grad_(True) for p in init]
+    for p, rp, gr in zip(ps, rps, gs):
+        p.grad = gr.clone()
+        rp.grad = gr.clone()
+    opt = FusedAdam(ps, lr=1e-2, weight_decay=0.1)
+    ref = torch.optim.AdamW(rps, lr=1e-2, weight_decay=0.1)
+    for _ in range(5):
+        opt.step()
+        ref.step()
+    for p, rp in zip(ps, rps):
+        assert rel_err(p, rp) < 1e-4
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm / RMSNorm (K4)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("cols", [1024, 4096, 16384])
+def test_layernorm_fwd_bwd(cols):
+    requires_ext()
+    from saturn_amd.ops.functional import fused_layer_norm
+
+    torch.manual_seed(0)
+    rows = 256
+    x32 = torch.randn(rows, cols, device="cuda", requires_grad=True)
+    w32 = torch.randn(cols, device="cuda", requires_grad=True)
+    b32 = torch.randn(cols, device="cuda", requires_grad=True)
+    y32 = torch.nn.functional.layer_norm(x32, (cols,), w32, b32)
+    dy = torch.randn_like(y32)
+    y32.backward(dy)
+
+    x = x32.detach().to(torch.bfloat16).requires_grad_(True)
+    w = w32.detach().to(torch.bfloat16).requires_grad_(True)
+    b = b32.detach().to(torch.bfloat16).requires_grad_(True)
+    y = fused_layer_norm(x, w, b)
+    y.backward(dy.to(torch.bfloat16))
+    assert rel_err(y, y32) < 2e-2
+    assert rel_err(x.grad, x32.grad) < 4e-2
+    assert rel_err(w.grad, w32.grad) < 4e-2
+    assert rel_err(b.grad, b32.grad) < 4e-2
+
+
+def test_rmsnorm_fwd_bwd():
+    requires_ext()
+    from saturn_amd.ops.functional import fused_rms_norm
+
+    torch.manual_seed(0)
+    rows, cols = 512, 4096
+    x32 = torch.randn(rows, cols, device="cuda", requires_grad=True)
+    w32 = torch.randn(cols, device="cuda", requires_grad=True)
+    y32 = x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-6) * w32
+    dy = torch.randn_like(y32)
+    y32.backward(dy)
+    x = x32.detach().to(torch.bfloat16).requires_grad_(True)
+    w = w32.detach().to(torch.bfloat16).requires_grad_(True)
+    y = fused_rms_norm(x, w)
+    y.backward(dy.to(torch.bfloat16))
+    assert rel_err(y, y32) < 2e-2
+    assert rel_err(x.grad, x32.grad) < 4e-2
+    assert rel_err(w.grad, w32.grad) < 4e-2
+
+
+# ---------------------------------------------------------------------------
+# Fused cross-entropy (K8)
+# ---------------------------------------------------------------------------
+def test_cross_entropy_fwd_bwd():
+    requires_ext()
+    from saturn_amd.ops.functional import fused_cross_entropy
+
+    torch.manual_seed(0)
+    B, T, V = 4, 128, 50400
+    logits32 = (torch.randn(B, T, V, device="cuda") * 4).requires_grad_(True)
+    targets = torch.randint(0, V, (B, T), device="cuda")
+    ref = torch.nn.functional.cross_entropy(
+        logits32[:, :-1].reshape(-1, V), targets[:, 1:].reshape(-1)
+    )
+    ref.backward()
+
+    lg = logits32.detach().to(torch.bfloat16).requires_grad_(True)
+    loss = fused_cross_entropy(lg, targets, shift=True)
+    loss.backward()
+    assert abs(loss.item() - ref.item()) / ref.item() < 2e-2
+    assert rel_err(lg.grad, logits32.grad) < 5e-2
+
+
+def test_cross_entropy_ignore_index():
+    requires_ext()
+    from saturn_amd.ops.functional import fused_cross_entropy
+
+    torch.manual_seed(0)
+    B, T, V = 2, 64, 1000
+    logits = torch.randn(B, T, V, device="cuda", dtype=torch.bfloat16)
+    targets = torch.randint(0, V, (B, T), device="cuda")
+    targets[:, ::2] = -100
+    loss = fused_cross_entropy(logits, targets, shift=True)
+    ref = torch.nn.functional.cross_entropy(
+        logits[:, :-1].float().reshape(-1, V),
+        targets[:, 1:].reshape(-1),
+        ignore_index=-100,
+    )
+    assert abs(loss.item() - ref.item()) / max(1e-6, ref.item()) < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# RoPE (K3)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("half_style", [False, True])
+def test_rope_matches_cpu(half_style):
+    requires_ext()
+    from saturn_amd.ops.functional import apply_rope, rope_tables
+
+    torch.manual_seed(0)
+    B, T, H, D = 2, 64, 4, 64
+    rot = 32
+    cos, sin = rope_tables(T, rot)
+    x = torch.randn(B, T, H, D)
+    y_cpu = apply_rope(x, cos, sin, half_style)
+    y_gpu = apply_rope(
+        x.cuda().to(torch.bfloat16), cos.cuda(), sin.cuda(), half_style
+    )
+    assert rel_err(y_gpu.cpu(), y_cpu) < 2e-2
+
+
+def test_rope_backward_is_inverse_rotation():
+    requires_ext()
+    from saturn_amd.ops.functional import apply_rope, rope_tables
+
+    torch.manual_seed(0)
+    B, T, H, D = 2, 32, 2, 32
+    cos, sin = rope_tables(T, D)
+    x32 = torch.randn(B, T, H, D, device="cuda", requires_grad=True)
+    # fp32 reference path on CPU via autograd
+    xc = x32.detach().cpu().requires_grad_(True)
+    y_ref = apply_rope(xc, cos, sin, False)
+    dy = torch.randn(B, T, H, D)
+    y_ref.backward(dy)
+    x = x32.detach().to(torch.bfloat16).requires_grad_(True)
+    y = apply_rope(x, cos.cuda(), sin.cuda(), False)
+    y.backward(dy.cuda().to(torch.bfloat16))
+    assert rel_err(x.grad.cpu(), xc.grad) < 2e-2
+
+
+# ---------------------------------------------------------------------------
+# Attention dispatch (math fallback or flash kernel, whichever is built)
+# ---------------------------------------------------------------------------
+def test_causal_attention_matches_fp32_math():
+    from saturn_amd.ops.functional import attention_math, causal_attention
+
+    torch.manual_seed(0)
+    B, H, T, D = 2, 4, 128, 128
+    q32 = torch.randn(B, H, T, D, device="cuda")
+    k32 = torch.randn(B, H, T, D, device="cuda")
+    v32 = torch.randn(B, H, T, D, device="cuda")
+    ref = attention_math(q32, k32, v32)
+    out = causal_attention(
+        q32.to(torch.bfloat16), k32.to(torch.bfloat16), v32.to(torch.bfloat16)
+    )
+    assert rel_err(out, ref) < 4e-2

@@ -1,0 +1,59 @@
+"""GPU integration: tiny GPT-J trains (loss decreases) on the fused stack,
+and the DDP executor's single-GPU path works end to end."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_gptj_tiny_loss_decreases():
+    from saturn_amd.models.gptj import GPTJConfig, GPTJForCausalLM, pretraining_loss
+    from saturn_amd.ops import require_ext
+    from saturn_amd.ops.optim import FusedAdam
+
+    require_ext()
+    torch.manual_seed(0)
+    cfg = GPTJConfig(n_layer=2, n_embd=512, n_head=4, n_ctx=128,
+                     vocab_size=2048, rotary_dim=32)
+    model = GPTJForCausalLM(cfg).to("cuda", torch.bfloat16)
+    opt = FusedAdam(model.parameters(), lr=3e-4)
+    x = torch.randint(0, cfg.vocab_size, (4, 128), device="cuda")
+    losses = []
+    for _ in range(20):
+        loss = pretraining_loss(model(x), x)
+        loss.backward()
+        opt.step()
+        for p in model.parameters():
+            p.grad = None
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, f"no learning: {losses[0]} -> {losses[-1]}"
+
+
+def test_bucketed_ddp_single_gpu_step():
+    from saturn_amd.models.gptj import GPTJConfig, GPTJForCausalLM, pretraining_loss
+    from saturn_amd.ops.optim import FusedSGD
+    from saturn_amd.parallel.ddp import BucketedDDP
+
+    torch.manual_seed(0)
+    cfg = GPTJConfig(n_layer=2, n_embd=256, n_head=4, n_ctx=64,
+                     vocab_size=1024, rotary_dim=16)
+    model = GPTJForCausalLM(cfg).to("cuda", torch.bfloat16)
+    ddp = BucketedDDP(model, bucket_mb=4)
+    opt = FusedSGD(model.parameters(), lr=1e-3)
+    x = torch.randint(0, cfg.vocab_size, (2, 64), device="cuda")
+    for _ in range(2):
+        loss = pretraining_loss(ddp(x), x)
+        loss.backward()
+        ddp.grad_sync()
+        opt.step()
+        ddp.zero_grad_buffers()
+    assert torch.isfinite(torch.tensor(float(loss)))
+
+
+def test_native_extension_is_loaded():
+    """Round-end check mirror: the in-tree .so must be the loaded one."""
+    import saturn_amd._C as C
+
+    assert "/saturn_amd/" in C.__file__, C.__file__
+    assert C.__file__.endswith(".so")
