@@ -195,8 +195,9 @@ class Task:
         optimizer moments at every interval boundary; we keep
         both under the same ``<name>.pt`` path.
         """
+        state = model if isinstance(model, dict) else model.state_dict()
         payload = {
-            "model": {k: v.cpu() for k, v in model.state_dict().items()},
+            "model": {k: v.cpu() for k, v in state.items()},
             "optimizer": optimizer.state_dict() if optimizer is not None else None,
             "extra": extra or {},
         }

@@ -25,6 +25,8 @@ class _NormFn(torch.autograd.Function):
         ext = require_ext()
         x = x.contiguous()
         y, mean, rstd = ext.norm_fwd(x, w, b, eps, rms)
+        if rms:
+            mean = rstd  # placeholder; RMS backward ignores it
         ctx.save_for_backward(x, w, mean, rstd)
         ctx.rms = rms
         ctx.has_b = b is not None
@@ -209,8 +211,9 @@ def apply_rope(
     """
     B, T, H, D = x.shape
     half = cos.shape[-1]
-    cos_bt = cos.unsqueeze(0).expand(B, T, half).contiguous()
-    sin_bt = sin.unsqueeze(0).expand(B, T, half).contiguous()
+    # model.to(dtype=bf16) converts registered buffers — force fp32 tables
+    cos_bt = cos.float().unsqueeze(0).expand(B, T, half).contiguous()
+    sin_bt = sin.float().unsqueeze(0).expand(B, T, half).contiguous()
     if x.is_cuda:
         return _RopeFn.apply(x, cos_bt, sin_bt, half_style)
     return _rope_cpu(x, cos_bt, sin_bt, half_style)
