@@ -1,0 +1,240 @@
+#include "hip/hip_runtime.h"
+// Fused causal flash attention forward for CDNA4.
+//
+// Replaces the reference's six-op attention chain (GPTJ.py:164-191:
+// matmul, where-mask, scale, softmax, dropout, matmul with a materialized
+// fp32 [T,T] score tensor) with one MFMA kernel using online softmax
+// (O(T) memory) and fp32 accumulation.
+//
+// v1 structure (correctness-first; the 8-wave swapped-QK^T structure of the
+// CDNA4 guide is the optimization target for later passes):
+//   grid  = (T/64, B*H); block = 256 threads = 4 waves
+//   each wave owns 16 q rows; the block shares K/V tiles of 32 keys staged
+//   in LDS; per kv-tile each wave computes S[16][32] with
+//   v_mfma_f32_16x16x32_bf16 (contraction over D in 32-slices), does the
+//   online-softmax update through wave-private LDS, and accumulates
+//   O[16][D] with PV MFMAs.
+//
+// Fragment layouts validated on MI355X hardware (tools/mfma_probe.hip):
+//   A[m][k]: lane l, slot j (0..7) -> A[l&15][(l>>4)*8 + j]
+//   B[k][n]: lane l, slot j        -> B[(l>>4)*8 + j][l&15]
+//   D[m][n]: lane l, reg r (0..3)  -> D[(l>>4)*4 + r][l&15]
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace samd {
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+constexpr int QBLK = 64;   // q rows per block
+constexpr int KBLK = 32;   // keys per kv tile
+constexpr int WROWS = 16;  // q rows per wave
+
+// LDS layout (per block), bf16 K/V tiles + per-wave fp32 scratch:
+//   K[32][D], V[32][D] (+8 bf16 row pad to break bank conflicts)
+//   per wave: P[16][32] bf16, m[16], l[16], alpha[16] fp32
+template <int D>
+struct AttnLds {
+  unsigned short k[KBLK][D + 8];
+  unsigned short v[KBLK][D + 8];
+  unsigned short p[4][WROWS][KBLK];
+  float s[4][WROWS][KBLK + 2];
+  float m[4][WROWS];
+  float l[4][WROWS];
+  float alpha[4][WROWS];
+};
+
+template <int D>
+__launch_bounds__(256, 2)
+__global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
+                                const unsigned short* __restrict__ K,
+                                const unsigned short* __restrict__ V,
+                                unsigned short* __restrict__ O,
+                                float* __restrict__ LSE, int T, int n_heads,
+                                float scale, int causal) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  AttnLds<D>& lds = *reinterpret_cast<AttnLds<D>*>(smem);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int q0_block = blockIdx.x * QBLK;        // block's first q row
+  const int q0 = q0_block + wid * WROWS;         // wave's first q row
+  const long bh = blockIdx.y;
+  const unsigned short* Qh = Q + bh * (long)T * D;
+  const unsigned short* Kh = K + bh * (long)T * D;
+  const unsigned short* Vh = V + bh * (long)T * D;
+
+  const int r = lane & 15;       // fragment row/col index
+  const int qg = lane >> 4;      // quadrant 0..3
+
+  // ---- load Q fragments to registers: frag_q[ds][j] = Q[q0 + r][ds*32 + qg*8 + j]
+  bf16x8 frag_q[D / 32];
+#pragma unroll
+  for (int ds = 0; ds < D / 32; ++ds) {
+    const unsigned short* src = Qh + (long)(q0 + r) * D + ds * 32 + qg * 8;
+    frag_q[ds] = *reinterpret_cast<const bf16x8*>(src);
+  }
+
+  // ---- O accumulators: otile[dt][reg] = O[(qg*4+reg)][dt*16 + r]
+  f32x4 otile[D / 16];
+#pragma unroll
+  for (int dt = 0; dt < D / 16; ++dt) otile[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  // ---- init softmax state
+  if (lane < WROWS) {
+    lds.m[wid][lane] = -INFINITY;
+    lds.l[wid][lane] = 0.f;
+  }
+
+  const int kv_end = causal ? (q0_block + QBLK) : T;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KBLK) {
+    // ---- stage K/V tile cooperatively (256 threads, coalesced rows)
+    __syncthreads();
+    {
+      // each thread copies (KBLK*D)/256 bf16 elements as 8-wide chunks
+      constexpr int CHUNKS = (KBLK * D) / (256 * 8);
+      for (int c = 0; c < CHUNKS; ++c) {
+        int idx = (c * 256 + threadIdx.x) * 8;
+        int row = idx / D, col = idx % D;
+        *reinterpret_cast<bf16x8*>(&lds.k[row][col]) =
+            *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
+        *reinterpret_cast<bf16x8*>(&lds.v[row][col]) =
+            *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
+      }
+    }
+    __syncthreads();
+
+    // ---- S[16][32] = scale * Q K^T  (two 16-col MFMA tiles)
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ds = 0; ds < D / 32; ++ds) {
+      bf16x8 bk0 = *reinterpret_cast<const bf16x8*>(&lds.k[r][ds * 32 + qg * 8]);
+      bf16x8 bk1 =
+          *reinterpret_cast<const bf16x8*>(&lds.k[16 + r][ds * 32 + qg * 8]);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_q[ds], bk0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_q[ds], bk1, acc1, 0, 0, 0);
+    }
+    // write S to wave-private LDS with causal mask
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int row = qg * 4 + reg;           // S row (q row = q0 + row)
+      int qrow = q0 + row;
+      float s0 = acc0[reg] * scale;
+      float s1 = acc1[reg] * scale;
+      if (causal) {
+        if (kv0 + r > qrow) s0 = -INFINITY;
+        if (kv0 + 16 + r > qrow) s1 = -INFINITY;
+      }
+      lds.s[wid][row][r] = s0;
+      lds.s[wid][row][16 + r] = s1;
+    }
+
+    // ---- online softmax (wave-private; lanes l, l+16, l+32, l+48 share row r)
+    {
+      float tm = -INFINITY;
+#pragma unroll
+      for (int c = 0; c < 8; ++c)
+        tm = fmaxf(tm, lds.s[wid][r][qg * 8 + c]);
+      tm = fmaxf(tm, __shfl_xor(tm, 16, 64));
+      tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
+      float m_old = lds.m[wid][r];
+      float m_new = fmaxf(m_old, tm);
+      float al = (m_old == -INFINITY) ? 0.f : __expf(m_old - m_new);
+      float ps = 0.f;
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        float sv = lds.s[wid][r][qg * 8 + c];
+        float pv = (sv == -INFINITY) ? 0.f : __expf(sv - m_new);
+        lds.p[wid][r][qg * 8 + c] = f2us(pv);
+        ps += pv;
+      }
+      ps += __shfl_xor(ps, 16, 64);
+      ps += __shfl_xor(ps, 32, 64);
+      if (lane < WROWS) {
+        lds.m[wid][r] = m_new;
+        lds.l[wid][r] = lds.l[wid][r] * al + ps;
+        lds.alpha[wid][r] = al;
+      }
+    }
+    // rescale O accumulators by this row's alpha
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      float al = lds.alpha[wid][qg * 4 + reg];
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) otile[dt][reg] *= al;
+    }
+
+    // ---- PV: O[16][D] += P[16][32] x V[32][D]
+    bf16x8 frag_p = *reinterpret_cast<const bf16x8*>(&lds.p[wid][r][qg * 8]);
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+      bf16x8 bv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        bv[j] = (short)lds.v[qg * 8 + j][dt * 16 + r];
+      otile[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_p, bv, otile[dt], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: O /= l, write bf16; LSE = m + log(l)
+  unsigned short* Oh = O + bh * (long)T * D;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    int row = qg * 4 + reg;
+    float linv = 1.f / lds.l[wid][row];
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt)
+      Oh[(long)(q0 + row) * D + dt * 16 + r] = f2us(otile[dt][reg] * linv);
+  }
+  if (lane < WROWS && LSE != nullptr) {
+    LSE[bh * (long)T + q0 + lane] =
+        lds.m[wid][lane] + __logf(lds.l[wid][lane]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host wrapper
+// ---------------------------------------------------------------------------
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 bool causal) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
+              "attn_fwd: bf16 only");
+  TORCH_CHECK(q.dim() == 4, "attn_fwd: [B, H, T, D]");
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
+            D = (int)q.size(3);
+  TORCH_CHECK(k.size(2) == T, "attn_fwd: q/k length mismatch");
+  TORCH_CHECK(T % QBLK == 0, "attn_fwd: T must be a multiple of 64");
+  TORCH_CHECK(D == 64 || D == 128 || D == 256, "attn_fwd: D in {64,128,256}");
+
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const float scale = 1.f / sqrtf((float)D);
+  dim3 grid(T / QBLK, B * H), block(256);
+
+#define LAUNCH(DD)                                                          \
+  do {                                                                      \
+    size_t shmem = sizeof(AttnLds<DD>);                                     \
+    hipLaunchKernelGGL((attn_fwd_kernel<DD>), grid, block, shmem,           \
+                       stream.stream(),                                     \
+                       reinterpret_cast<const unsigned short*>(q.data_ptr()), \
+                       reinterpret_cast<const unsigned short*>(k.data_ptr()), \
+                       reinterpret_cast<const unsigned short*>(v.data_ptr()), \
+                       reinterpret_cast<unsigned short*>(o.data_ptr()),     \
+                       lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0); \
+  } while (0)
+
+  if (D == 64) LAUNCH(64);
+  else if (D == 128) LAUNCH(128);
+  else LAUNCH(256);
+#undef LAUNCH
+  return {o, lse};
+}
+
+}  // namespace samd

@@ -23,6 +23,8 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                   at::Tensor dloss, int64_t Tr, int64_t ignore_index);
 void rope_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
                 bool half_style, bool backward);
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 bool causal);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -34,4 +36,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &samd::ce_fwd, "fused cross-entropy forward");
   m.def("ce_bwd", &samd::ce_bwd, "fused cross-entropy backward");
   m.def("rope_apply", &samd::rope_apply, "fused rotary embedding (in-place)");
+  m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
 }

@@ -30,16 +30,46 @@ class _FlashFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, do):
-        ext = require_ext()
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = ext.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.causal)
-        return dq, dk, dv, None
+        ext = require_ext()
+        if hasattr(ext, "attn_bwd"):
+            dq, dk, dv = ext.attn_bwd(
+                do.contiguous(), q, k, v, o, lse, ctx.causal
+            )
+            return dq, dk, dv, None
+        # Analytic FA2 backward composed from rocBLAS GEMMs (recompute P
+        # from the saved LSE; O(T^2) transient, fp32 math).  The fused HIP
+        # backward kernel replaces this path when built.
+        scale = 1.0 / (q.shape[-1] ** 0.5)
+        qf, kf, vf = q.float(), k.float(), v.float()
+        dof, of = do.float(), o.float()
+        s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+        if ctx.causal:
+            T = q.shape[-2]
+            mask = torch.ones(T, T, dtype=torch.bool, device=q.device).tril()
+            s = s.masked_fill(~mask, float("-inf"))
+        p = torch.exp(s - lse.unsqueeze(-1))
+        dv = torch.matmul(p.transpose(-1, -2), dof)
+        dp = torch.matmul(dof, vf.transpose(-1, -2))
+        delta = (dof * of).sum(-1, keepdim=True)
+        ds = p * (dp - delta) * scale
+        dq = torch.matmul(ds, kf)
+        dk = torch.matmul(ds.transpose(-1, -2), qf)
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
+
+
+def _kernel_supported(q) -> bool:
+    return (
+        q.dtype == torch.bfloat16
+        and q.shape[-2] % 64 == 0
+        and q.shape[-1] in (64, 128, 256)
+    )
 
 
 def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
     """q,k,v: [B, H, T, D] bf16/fp16 contiguous (GQA: H_kv may divide H)."""
     ext = require_ext()
-    if hasattr(ext, "attn_fwd"):
+    if hasattr(ext, "attn_fwd") and _kernel_supported(q):
         if k.shape[1] != q.shape[1]:  # GQA: expand kv heads
             rep = q.shape[1] // k.shape[1]
             k = k.repeat_interleave(rep, dim=1)

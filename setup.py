@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "layernorm.hip"),
         os.path.join(CSRC, "cross_entropy.hip"),
         os.path.join(CSRC, "rope.hip"),
+        os.path.join(CSRC, "attention.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -227,3 +227,54 @@ def test_causal_attention_matches_fp32_math():
         q32.to(torch.bfloat16), k32.to(torch.bfloat16), v32.to(torch.bfloat16)
     )
     assert rel_err(out, ref) < 4e-2
+
+
+def test_flash_attention_fwd_vs_fp32(
+):
+    from saturn_amd.ops import require_ext
+    from saturn_amd.ops.flash import flash_attention
+    from saturn_amd.ops.functional import attention_math
+
+    ext = require_ext()
+    if not hasattr(ext, "attn_fwd"):
+        pytest.skip("attn_fwd not built")
+    torch.manual_seed(0)
+    for (B, H, T, D) in [(2, 4, 128, 64), (1, 2, 256, 128), (1, 2, 128, 256)]:
+        q32 = torch.randn(B, H, T, D, device="cuda")
+        k32 = torch.randn(B, H, T, D, device="cuda")
+        v32 = torch.randn(B, H, T, D, device="cuda")
+        ref = attention_math(q32, k32, v32, causal=True)
+        out = flash_attention(
+            q32.to(torch.bfloat16).contiguous(),
+            k32.to(torch.bfloat16).contiguous(),
+            v32.to(torch.bfloat16).contiguous(),
+        )
+        e = rel_err(out, ref)
+        assert e < 4e-2, f"shape {(B,H,T,D)}: rel err {e}"
+
+
+def test_flash_attention_bwd_vs_fp32():
+    from saturn_amd.ops import require_ext
+    from saturn_amd.ops.flash import flash_attention
+    from saturn_amd.ops.functional import attention_math
+
+    ext = require_ext()
+    if not hasattr(ext, "attn_fwd"):
+        pytest.skip("attn_fwd not built")
+    torch.manual_seed(0)
+    B, H, T, D = 2, 2, 128, 128
+    q32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    k32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    v32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    ref = attention_math(q32, k32, v32, causal=True)
+    do = torch.randn_like(ref)
+    ref.backward(do)
+
+    q = q32.detach().to(torch.bfloat16).requires_grad_(True)
+    k = k32.detach().to(torch.bfloat16).requires_grad_(True)
+    v = v32.detach().to(torch.bfloat16).requires_grad_(True)
+    out = flash_attention(q, k, v)
+    out.backward(do.to(torch.bfloat16))
+    assert rel_err(q.grad, q32.grad) < 5e-2
+    assert rel_err(k.grad, k32.grad) < 5e-2
+    assert rel_err(v.grad, v32.grad) < 5e-2
