@@ -1,5 +1,6 @@
 from .mlp import MLP, get_mlp_dataloader, get_mlp_model, mse_loss
 from .gpt2 import GPT2ForCausalLM, get_gpt2_model, gpt2_loss
+from .llama import LlamaForCausalLM, get_llama_model, llama_loss
 from .gptj import (
     GPTJForCausalLM,
     get_gptj_model,
@@ -19,4 +20,7 @@ __all__ = [
     "get_gptj_model",
     "make_token_dataloader",
     "pretraining_loss",
+    "LlamaForCausalLM",
+    "get_llama_model",
+    "llama_loss",
 ]

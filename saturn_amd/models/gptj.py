@@ -142,6 +142,34 @@ def pretraining_loss(logits, labels):
     return fused_cross_entropy(logits, labels, shift=True)
 
 
+class _EmbedStage(nn.Module):
+    def __init__(self, wte):
+        super().__init__()
+        self.wte = wte
+
+    def forward(self, input_ids):
+        return self.wte(input_ids)
+
+
+class _HeadStage(nn.Module):
+    def __init__(self, ln_f, lm_head):
+        super().__init__()
+        self.ln_f = ln_f
+        self.lm_head = lm_head
+
+    def forward(self, x):
+        return self.lm_head(self.ln_f(x))
+
+
+def as_sequential(model: GPTJForCausalLM) -> nn.Sequential:
+    """Flatten into an nn.Sequential for the pipeline executor (parameters
+    are shared with ``model``; the reference flattens GPT-J the same way for
+    torchgpipe, GPTJ.py:511-521)."""
+    return nn.Sequential(
+        _EmbedStage(model.wte), *model.h, _HeadStage(model.ln_f, model.lm_head)
+    )
+
+
 # ---------------------------------------------------------------------------
 # Factories (Task contract: picklable top-level callables)
 # ---------------------------------------------------------------------------

@@ -238,9 +238,13 @@ def attention_math(q, k, v, causal: bool = True) -> torch.Tensor:
 
 def causal_attention(q, k, v) -> torch.Tensor:
     """Dispatch: fused CDNA4 flash kernel on GPU (when built), math path on
-    CPU."""
+    CPU.  Handles GQA (k/v with fewer heads) by expansion."""
     if q.is_cuda:
         from saturn_amd.ops import flash  # local import: optional kernel
 
         return flash.flash_attention(q, k, v, causal=True)
+    if k.shape[1] != q.shape[1]:
+        rep = q.shape[1] // k.shape[1]
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
     return attention_math(q, k, v, causal=True)
