@@ -238,3 +238,254 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 }
 
 }  // namespace samd
+
+// ===========================================================================
+// Backward (FA2 scheme): grid over kv tiles; each block owns 64 keys
+// (16 per wave) and loops over q tiles of 32 rows, recomputing P from the
+// saved LSE.  dK/dV accumulate in registers across the q loop; dQ partials
+// go to a fp32 buffer with atomics (summed across kv blocks).
+//   dV = P^T dO;  dP^T = V dO^T;  dS^T = P^T o (dP^T - delta) * scale;
+//   dK = dS^T Q;  dQ = dS K;  delta = rowsum(dO o O) (computed by caller).
+// ===========================================================================
+
+namespace samd {
+
+namespace bwd {
+constexpr int KB = 64;   // keys per block
+constexpr int QT = 32;   // q rows per tile
+}
+
+template <int D>
+struct AttnBwdLds {
+  unsigned short kt[bwd::KB][D + 8];
+  unsigned short vt[bwd::KB][D + 8];
+  unsigned short qt[bwd::QT][D + 8];
+  unsigned short dot[bwd::QT][D + 8];
+  unsigned short pt[4][16][bwd::QT + 8];   // per-wave P^T tile
+  unsigned short dst[bwd::KB][bwd::QT + 8]; // shared dS^T (bf16, scaled)
+  float lse_t[bwd::QT];
+  float delta_t[bwd::QT];
+};
+
+template <int D>
+__launch_bounds__(256, 1)
+__global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
+                                const unsigned short* __restrict__ Q,
+                                const unsigned short* __restrict__ K,
+                                const unsigned short* __restrict__ V,
+                                const float* __restrict__ LSE,
+                                const float* __restrict__ DELTA,
+                                float* __restrict__ dQ,
+                                unsigned short* __restrict__ dK,
+                                unsigned short* __restrict__ dV, int T,
+                                float scale, int causal) {
+  using namespace bwd;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  AttnBwdLds<D>& lds = *reinterpret_cast<AttnBwdLds<D>*>(smem);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int r = lane & 15;
+  const int qg = lane >> 4;
+  const int kv0 = blockIdx.x * KB;
+  const long bh = blockIdx.y;
+  const unsigned short* Qh = Q + bh * (long)T * D;
+  const unsigned short* Kh = K + bh * (long)T * D;
+  const unsigned short* Vh = V + bh * (long)T * D;
+  const unsigned short* dOh = dO + bh * (long)T * D;
+  const float* lse_h = LSE + bh * (long)T;
+  const float* del_h = DELTA + bh * (long)T;
+
+  // ---- stage K/V tiles once (they live for the whole block)
+  {
+    constexpr int CHUNKS = (KB * D) / (256 * 8);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      int idx = (c * 256 + threadIdx.x) * 8;
+      int row = idx / D, col = idx % D;
+      *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
+          *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
+      *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
+          *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
+    }
+  }
+
+  f32x4 acc_dv[D / 16], acc_dk[D / 16];
+#pragma unroll
+  for (int dt = 0; dt < D / 16; ++dt) {
+    acc_dv[dt] = {0.f, 0.f, 0.f, 0.f};
+    acc_dk[dt] = {0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int q_start = causal ? kv0 : 0;
+  for (int q0 = q_start; q0 < T; q0 += QT) {
+    // ---- stage Q / dO tiles + lse/delta
+    __syncthreads();
+    {
+      constexpr int CHUNKS = (QT * D) / (256 * 8);
+#pragma unroll
+      for (int c = 0; c < CHUNKS; ++c) {
+        int idx = (c * 256 + threadIdx.x) * 8;
+        int row = idx / D, col = idx % D;
+        *reinterpret_cast<bf16x8*>(&lds.qt[row][col]) =
+            *reinterpret_cast<const bf16x8*>(Qh + (long)(q0 + row) * D + col);
+        *reinterpret_cast<bf16x8*>(&lds.dot[row][col]) =
+            *reinterpret_cast<const bf16x8*>(dOh + (long)(q0 + row) * D + col);
+      }
+      if (threadIdx.x < QT) {
+        lds.lse_t[threadIdx.x] = lse_h[q0 + threadIdx.x];
+        lds.delta_t[threadIdx.x] = del_h[q0 + threadIdx.x];
+      }
+    }
+    __syncthreads();
+
+    // ---- per 16-q subtile: St, P^T, dPt, dS^T
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      f32x4 st = {0.f, 0.f, 0.f, 0.f};
+      f32x4 dpt = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ds = 0; ds < D / 32; ++ds) {
+        bf16x8 ak = *reinterpret_cast<const bf16x8*>(
+            &lds.kt[wid * 16 + r][ds * 32 + qg * 8]);
+        bf16x8 av = *reinterpret_cast<const bf16x8*>(
+            &lds.vt[wid * 16 + r][ds * 32 + qg * 8]);
+        bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+            &lds.qt[n * 16 + r][ds * 32 + qg * 8]);
+        bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
+            &lds.dot[n * 16 + r][ds * 32 + qg * 8]);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, bq, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bdo, dpt, 0, 0, 0);
+      }
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int key_loc = qg * 4 + reg;           // within wave's 16 keys
+        const int key_glob = kv0 + wid * 16 + key_loc;
+        const int q_glob = q0 + n * 16 + r;
+        float p = 0.f;
+        if (!causal || q_glob >= key_glob) {
+          p = __expf(scale * st[reg] - lds.lse_t[n * 16 + r]);
+        }
+        const float dsv = p * (dpt[reg] - lds.delta_t[n * 16 + r]) * scale;
+        lds.pt[wid][key_loc][n * 16 + r] = f2us(p);
+        lds.dst[wid * 16 + key_loc][n * 16 + r] = f2us(dsv);
+      }
+    }
+
+    // ---- accumulate dV, dK over this q tile (contraction over 32 q)
+    {
+      bf16x8 ap = *reinterpret_cast<const bf16x8*>(&lds.pt[wid][r][qg * 8]);
+      bf16x8 adst =
+          *reinterpret_cast<const bf16x8*>(&lds.dst[wid * 16 + r][qg * 8]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        bf16x8 bdo, bq;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          bdo[j] = (short)lds.dot[qg * 8 + j][dt * 16 + r];
+          bq[j] = (short)lds.qt[qg * 8 + j][dt * 16 + r];
+        }
+        acc_dv[dt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bdo, acc_dv[dt], 0, 0, 0);
+        acc_dk[dt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bq, acc_dk[dt], 0, 0, 0);
+      }
+    }
+
+    // ---- dQ partials: wave w handles q-subtile (w&1), d-half (w>>1)
+    __syncthreads();
+    {
+      const int n = wid & 1;
+      const int h = wid >> 1;
+#pragma unroll
+      for (int dt2 = 0; dt2 < D / 32; ++dt2) {
+        const int dt = h * (D / 32) + dt2;
+        f32x4 acc_dq = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kt2 = 0; kt2 < 2; ++kt2) {
+          bf16x8 adst, bk;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][n * 16 + r];
+            bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
+          }
+          // A[q row][key k]: dst is [key][q]; the transposed read above
+          // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
+          acc_dq = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bk, acc_dq, 0, 0, 0);
+        }
+#pragma unroll
+        for (int reg = 0; reg < 4; ++reg) {
+          const int q_glob = q0 + n * 16 + qg * 4 + reg;
+          atomicAdd(&dQ[(bh * (long)T + q_glob) * D + dt * 16 + r],
+                    acc_dq[reg]);
+        }
+      }
+    }
+  }
+
+  // ---- epilogue: write dK, dV (bf16)
+  unsigned short* dKh = dK + bh * (long)T * D;
+  unsigned short* dVh = dV + bh * (long)T * D;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int key_glob = kv0 + wid * 16 + qg * 4 + reg;
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+      dKh[(long)key_glob * D + dt * 16 + r] = f2us(acc_dk[dt][reg]);
+      dVh[(long)key_glob * D + dt * 16 + r] = f2us(acc_dv[dt][reg]);
+    }
+  }
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                 at::Tensor v, at::Tensor o, at::Tensor lse,
+                                 bool causal) {
+  using namespace bwd;
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
+            D = (int)q.size(3);
+  TORCH_CHECK(T % KB == 0, "attn_bwd: T must be a multiple of 64");
+  TORCH_CHECK(D == 64 || D == 128 || D == 256);
+  TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous() && o.is_contiguous());
+
+  auto delta = (dout.to(at::kFloat) * o.to(at::kFloat)).sum(-1).contiguous();
+  auto dq_f32 = at::zeros({B, H, T, D}, q.options().dtype(at::kFloat));
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto stream = at::hip::getCurrentHIPStream();
+  const float scale = 1.f / sqrtf((float)D);
+  dim3 grid(T / KB, B * H), block(256);
+
+#define LAUNCH_B(DD)                                                         \
+  do {                                                                       \
+    size_t shmem = sizeof(AttnBwdLds<DD>);                                   \
+    static bool attr_set_##DD = [] {                                         \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD>),               \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
+          (int)sizeof(AttnBwdLds<DD>));                                      \
+      return true;                                                           \
+    }();                                                                     \
+    (void)attr_set_##DD;                                                     \
+    hipLaunchKernelGGL((attn_bwd_kernel<DD>), grid, block, shmem,            \
+                       stream.stream(),                                      \
+                       reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
+                       reinterpret_cast<const unsigned short*>(q.data_ptr()),  \
+                       reinterpret_cast<const unsigned short*>(k.data_ptr()),  \
+                       reinterpret_cast<const unsigned short*>(v.data_ptr()),  \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),       \
+                       dq_f32.data_ptr<float>(),                             \
+                       reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
+                       reinterpret_cast<unsigned short*>(dv.data_ptr()), T,  \
+                       scale, causal ? 1 : 0);                               \
+  } while (0)
+
+  if (D == 64) LAUNCH_B(64);
+  else if (D == 128) LAUNCH_B(128);
+  else LAUNCH_B(256);
+#undef LAUNCH_B
+  return {dq_f32.to(at::kBFloat16), dk, dv};
+}
+
+}  // namespace samd

@@ -25,6 +25,9 @@ void rope_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
                 bool half_style, bool backward);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  bool causal);
+std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
+                                 at::Tensor v, at::Tensor o, at::Tensor lse,
+                                 bool causal);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -37,4 +40,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &samd::ce_bwd, "fused cross-entropy backward");
   m.def("rope_apply", &samd::rope_apply, "fused rotary embedding (in-place)");
   m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
+  m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward");
 }

@@ -121,8 +121,12 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
   const int B = (int)logits.size(0);
   const int V = (int)logits.size(2);
   const int n_rows = (int)(B * Tr);
-  // Gradient for the full [B, T, V] tensor; positions beyond Tr get zero.
-  auto dlogits = at::zeros_like(logits);
+  const int T = (int)logits.size(1);
+  // Gradient for the full [B, T, V] tensor; the kernel writes rows < Tr,
+  // only the (T - Tr) tail rows need zeroing (avoids a full 2x write of
+  // the 400+ MB grad tensor that zeros_like would cost).
+  auto dlogits = at::empty_like(logits);
+  if (T > Tr) dlogits.narrow(1, Tr, T - Tr).zero_();
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(std::min(n_rows, 2048)), block(CE_BLOCK);
   AT_DISPATCH_FLOATING_TYPES_AND2(

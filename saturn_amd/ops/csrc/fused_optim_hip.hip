@@ -1,12 +1,11 @@
 #include "hip/hip_runtime.h"
 // Fused multi-tensor SGD / Adam for CDNA4.
 //
-// One launch updates up to MAX_T parameter tensors: the pointer table rides
-// in the kernel-argument block (no host->device metadata copies), each
-// thread grid-strides the concatenated element space and resolves its
-// tensor by an unrolled scan of the cumulative-size table.  HBM-bound:
-// bf16 paths load/store 2 elements per instruction via ushort2-style
-// packing; grids are capped and grid-stride per guide Guideline 11.
+// One launch updates up to MAX_T parameter tensors; the pointer table rides
+// in the kernel-argument block (no host->device metadata copies).  The
+// element space is walked in 8-element groups (16 B of bf16 per lane —
+// guide G13: scalar bf16 loads cost ~2-2.5x) with per-tensor scalar tails;
+// grids are capped and grid-stride per guide Guideline 11.
 //
 // Replaces the reference's stock torch.optim.SGD step
 // (examples/wikitext103/simple-verification.py:59) with the fused kernel
@@ -21,19 +20,38 @@
 namespace samd {
 
 constexpr int MAX_T = 32;
+constexpr int GRP = 8;  // elements per group (16 B of bf16)
+
+typedef __attribute__((ext_vector_type(8))) short short8v_;
+typedef __attribute__((ext_vector_type(4))) float float4v_;
 
 struct SgdArgs {
   void* p[MAX_T];
   void* g[MAX_T];
-  void* m[MAX_T];  // momentum buffers (nullptr if momentum==0)
-  long cum[MAX_T + 1];  // cumulative numels
+  void* m[MAX_T];       // momentum buffers (fp32; nullptr set if unused)
+  long cum[MAX_T + 1];  // cumulative GROUP counts
+  long numel[MAX_T];
   int n_tensors;
   float lr, momentum, weight_decay;
 };
 
 template <typename T, bool HAS_M>
+__device__ __forceinline__ void sgd_elem(T* p, const T* g, float* m, long j,
+                                         const SgdArgs& a) {
+  float pv = toF<T>(p[j]);
+  float gv = toF<T>(g[j]);
+  if (a.weight_decay != 0.f) gv += a.weight_decay * pv;
+  if (HAS_M) {
+    float mv = m[j] * a.momentum + gv;
+    m[j] = mv;
+    gv = mv;
+  }
+  p[j] = fromF<T>(pv - a.lr * gv);
+}
+
+template <typename T, bool HAS_M>
 __global__ void fused_sgd_kernel(SgdArgs a) {
-  const long total = a.cum[a.n_tensors];
+  const long total = a.cum[a.n_tensors];  // total groups
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += stride) {
@@ -41,19 +59,50 @@ __global__ void fused_sgd_kernel(SgdArgs a) {
 #pragma unroll
     for (int k = 0; k < MAX_T; ++k)
       if (k < a.n_tensors && i >= a.cum[k + 1]) t = k + 1;
-    const long j = i - a.cum[t];
+    const long grp = i - a.cum[t];
+    const long base = grp * GRP;
     T* p = reinterpret_cast<T*>(a.p[t]);
     const T* g = reinterpret_cast<const T*>(a.g[t]);
-    float pv = (float)p[j];
-    float gv = (float)g[j];
-    if (a.weight_decay != 0.f) gv += a.weight_decay * pv;
-    if (HAS_M) {
-      float* m = reinterpret_cast<float*>(a.m[t]);
-      float mv = m[j] * a.momentum + gv;
-      m[j] = mv;
-      gv = mv;
+    float* m = HAS_M ? reinterpret_cast<float*>(a.m[t]) : nullptr;
+    const long n = a.numel[t];
+    if constexpr (sizeof(T) == 2) {
+      if (base + GRP <= n) {
+        // vector path: 8 x 16-bit in one 16-B load/store
+        short8v_ pv = *reinterpret_cast<short8v_*>(p + base);
+        short8v_ gv = *reinterpret_cast<const short8v_*>(g + base);
+        float4v_ m0, m1;
+        if (HAS_M) {
+          m0 = *reinterpret_cast<float4v_*>(m + base);
+          m1 = *reinterpret_cast<float4v_*>(m + base + 4);
+        }
+#pragma unroll
+        for (int e = 0; e < GRP; ++e) {
+          T pe, ge;
+          pe.x = (unsigned short)pv[e];
+          ge.x = (unsigned short)gv[e];
+          float pf = toF<T>(pe);
+          float gf = toF<T>(ge);
+          if (a.weight_decay != 0.f) gf += a.weight_decay * pf;
+          if (HAS_M) {
+            float mv = (e < 4 ? m0[e] : m1[e - 4]) * a.momentum + gf;
+            if (e < 4) m0[e] = mv; else m1[e - 4] = mv;
+            gf = mv;
+          }
+          T out = fromF<T>(pf - a.lr * gf);
+          pv[e] = (short)out.x;
+        }
+        *reinterpret_cast<short8v_*>(p + base) = pv;
+        if (HAS_M) {
+          *reinterpret_cast<float4v_*>(m + base) = m0;
+          *reinterpret_cast<float4v_*>(m + base + 4) = m1;
+        }
+      } else {
+        for (long j = base; j < n; ++j) sgd_elem<T, HAS_M>(p, g, m, j, a);
+      }
+    } else {
+      const long end = (base + GRP <= n) ? base + GRP : n;
+      for (long j = base; j < end; ++j) sgd_elem<T, HAS_M>(p, g, m, j, a);
     }
-    p[j] = (T)(pv - a.lr * gv);
   }
 }
 
@@ -63,9 +112,26 @@ struct AdamArgs {
   void* m[MAX_T];
   void* v[MAX_T];
   long cum[MAX_T + 1];
+  long numel[MAX_T];
   int n_tensors;
   float lr, beta1, beta2, eps, weight_decay, bc1, bc2;
 };
+
+template <typename T>
+__device__ __forceinline__ void adam_elem(T* p, const T* g, float* m, float* v,
+                                          long j, const AdamArgs& a,
+                                          float inv_bc1, float inv_bc2) {
+  float gv = toF<T>(g[j]);
+  float mv = a.beta1 * m[j] + (1.f - a.beta1) * gv;
+  float vv = a.beta2 * v[j] + (1.f - a.beta2) * gv * gv;
+  m[j] = mv;
+  v[j] = vv;
+  float denom = sqrtf(vv * inv_bc2) + a.eps;
+  float upd = (mv * inv_bc1) / denom;
+  float pv = toF<T>(p[j]);
+  if (a.weight_decay != 0.f) pv *= (1.f - a.lr * a.weight_decay);
+  p[j] = fromF<T>(pv - a.lr * upd);
+}
 
 template <typename T>
 __global__ void fused_adam_kernel(AdamArgs a) {
@@ -79,29 +145,62 @@ __global__ void fused_adam_kernel(AdamArgs a) {
 #pragma unroll
     for (int k = 0; k < MAX_T; ++k)
       if (k < a.n_tensors && i >= a.cum[k + 1]) t = k + 1;
-    const long j = i - a.cum[t];
+    const long grp = i - a.cum[t];
+    const long base = grp * GRP;
     T* p = reinterpret_cast<T*>(a.p[t]);
     const T* g = reinterpret_cast<const T*>(a.g[t]);
     float* m = reinterpret_cast<float*>(a.m[t]);
     float* v = reinterpret_cast<float*>(a.v[t]);
-    float gv = (float)g[j];
-    float mv = a.beta1 * m[j] + (1.f - a.beta1) * gv;
-    float vv = a.beta2 * v[j] + (1.f - a.beta2) * gv * gv;
-    m[j] = mv;
-    v[j] = vv;
-    float denom = sqrtf(vv * inv_bc2) + a.eps;
-    float upd = (mv * inv_bc1) / denom;
-    float pv = (float)p[j];
-    if (a.weight_decay != 0.f) pv *= (1.f - a.lr * a.weight_decay);
-    p[j] = (T)(pv - a.lr * upd);
+    const long n = a.numel[t];
+    if constexpr (sizeof(T) == 2) {
+      if (base + GRP <= n) {
+        // 16-B loads on p/g; moments are fp32 (2x16-B each)
+        short8v_ pv = *reinterpret_cast<short8v_*>(p + base);
+        short8v_ gv = *reinterpret_cast<const short8v_*>(g + base);
+        float4v_ m0 = *reinterpret_cast<float4v_*>(m + base);
+        float4v_ m1 = *reinterpret_cast<float4v_*>(m + base + 4);
+        float4v_ v0 = *reinterpret_cast<float4v_*>(v + base);
+        float4v_ v1 = *reinterpret_cast<float4v_*>(v + base + 4);
+#pragma unroll
+        for (int e = 0; e < GRP; ++e) {
+          T ge;
+          ge.x = (unsigned short)gv[e];
+          float gf = toF<T>(ge);
+          float mv = a.beta1 * (e < 4 ? m0[e] : m1[e - 4]) + (1.f - a.beta1) * gf;
+          float vv = a.beta2 * (e < 4 ? v0[e] : v1[e - 4]) + (1.f - a.beta2) * gf * gf;
+          if (e < 4) { m0[e] = mv; v0[e] = vv; } else { m1[e-4] = mv; v1[e-4] = vv; }
+          float denom = sqrtf(vv * inv_bc2) + a.eps;
+          float upd = (mv * inv_bc1) / denom;
+          T pe;
+          pe.x = (unsigned short)pv[e];
+          float pf = toF<T>(pe);
+          if (a.weight_decay != 0.f) pf *= (1.f - a.lr * a.weight_decay);
+          T out = fromF<T>(pf - a.lr * upd);
+          pv[e] = (short)out.x;
+        }
+        *reinterpret_cast<short8v_*>(p + base) = pv;
+        *reinterpret_cast<float4v_*>(m + base) = m0;
+        *reinterpret_cast<float4v_*>(m + base + 4) = m1;
+        *reinterpret_cast<float4v_*>(v + base) = v0;
+        *reinterpret_cast<float4v_*>(v + base + 4) = v1;
+      } else {
+        for (long j = base; j < n; ++j)
+          adam_elem<T>(p, g, m, v, j, a, inv_bc1, inv_bc2);
+      }
+    } else {
+      const long end = (base + GRP <= n) ? base + GRP : n;
+      for (long j = base; j < end; ++j)
+        adam_elem<T>(p, g, m, v, j, a, inv_bc1, inv_bc2);
+    }
   }
 }
 
 static int grid_for(long total, int block) {
   long g = (total + block - 1) / block;
-  // 256 CUs x 8 blocks/CU cap, grid-stride the rest (guide G11).
-  return (int)std::min<long>(g, 2048);
+  return (int)std::min<long>(std::max<long>(g, 1), 2048);
 }
+
+static long groups_of(long numel) { return (numel + GRP - 1) / GRP; }
 
 template <typename scalar_t>
 static void sgd_launch(std::vector<at::Tensor>& params,
@@ -122,7 +221,8 @@ static void sgd_launch(std::vector<at::Tensor>& params,
       a.p[k] = params[base + k].data_ptr();
       a.g[k] = grads[base + k].data_ptr();
       a.m[k] = has_m ? moms[base + k].data_ptr() : nullptr;
-      cum += params[base + k].numel();
+      a.numel[k] = params[base + k].numel();
+      cum += groups_of(a.numel[k]);
       a.cum[k + 1] = cum;
     }
     const int block = 256;
@@ -152,7 +252,7 @@ void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   switch (params[0].scalar_type()) {
     case at::kBFloat16:
       sgd_launch<c10::BFloat16>(params, grads, moms, lr, momentum,
-                                 weight_decay);
+                                weight_decay);
       break;
     case at::kFloat:
       sgd_launch<float>(params, grads, moms, lr, momentum, weight_decay);
@@ -190,7 +290,8 @@ static void adam_launch(std::vector<at::Tensor>& params,
       a.g[k] = grads[base + k].data_ptr();
       a.m[k] = ms[base + k].data_ptr();
       a.v[k] = vs[base + k].data_ptr();
-      cum += params[base + k].numel();
+      a.numel[k] = params[base + k].numel();
+      cum += groups_of(a.numel[k]);
       a.cum[k + 1] = cum;
     }
     const int block = 256;
@@ -216,7 +317,7 @@ void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
   switch (params[0].scalar_type()) {
     case at::kBFloat16:
       adam_launch<c10::BFloat16>(params, grads, ms, vs, lr, beta1, beta2, eps,
-                                  weight_decay, bc1, bc2);
+                                 weight_decay, bc1, bc2);
       break;
     case at::kFloat:
       adam_launch<float>(params, grads, ms, vs, lr, beta1, beta2, eps,
