@@ -75,7 +75,18 @@ def _ddp_worker(
             except Exception:
                 pass  # optimizer class may have changed between intervals
 
-        it = task.get_iterator() if not trial else task.get_fresh_iterator()
+        def sharded(base_iter):
+            """Round-robin batch sharding across ranks (capability parity
+            with the reference's DistributedSampler dataloader,
+            DDP.py:117-144; resume position under changed world size is
+            approximate there too — SURVEY quirk #11)."""
+            for i, b in enumerate(base_iter):
+                if i % world == rank:
+                    yield b
+
+        it = sharded(
+            task.get_iterator() if not trial else task.get_fresh_iterator()
+        )
 
         def step(batch) -> None:
             x, y = batch
@@ -93,7 +104,7 @@ def _ddp_worker(
             try:
                 return next(it)
             except StopIteration:
-                it = task.get_fresh_iterator()
+                it = sharded(task.get_fresh_iterator())
                 return next(it)
 
         result = None

@@ -66,16 +66,11 @@ class GPTJAttention(nn.Module):
         k = self.k_proj(x).view(B, T, H, D)
         v = self.v_proj(x).view(B, T, H, D)
         cos, sin = self.rope_cos[:T], self.rope_sin[:T]
-        if self.rotary_dim < D:
-            q = torch.cat(
-                [apply_rope(q[..., : self.rotary_dim].contiguous(), cos, sin),
-                 q[..., self.rotary_dim :]], dim=-1)
-            k = torch.cat(
-                [apply_rope(k[..., : self.rotary_dim].contiguous(), cos, sin),
-                 k[..., self.rotary_dim :]], dim=-1)
-        else:
-            q = apply_rope(q, cos, sin)
-            k = apply_rope(k, cos, sin)
+        # partial rotary: the kernel rotates the first rotary_dim dims in
+        # place and passes the rest through — no split+cat (the reference
+        # materializes both halves and concatenates, GPTJ.py:255-259)
+        q = apply_rope(q, cos, sin)
+        k = apply_rope(k, cos, sin)
         q = q.transpose(1, 2)  # [B, H, T, D]
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)

@@ -215,19 +215,31 @@ class Zero3Model(nn.Module):
         self._install_hooks()
 
     # -- hooks -------------------------------------------------------------
+    @staticmethod
+    def _triggers(u: _Unit) -> List[nn.Module]:
+        """Forward-invoked modules of a unit.  A container unit (e.g. a
+        ModuleList grouping several blocks for the spill executor's
+        partitions) never runs its own forward — hook its children."""
+        m = u.module
+        if isinstance(m, (nn.ModuleList, nn.ModuleDict)):
+            return list(m.children())
+        return [m]
+
     def _install_hooks(self) -> None:
         for u in self.units:
             if u.module is self.model:
                 continue
-            u.module.register_forward_pre_hook(self._fwd_pre(u))
-            u.module.register_forward_hook(self._fwd_post(u))
-            u.module.register_full_backward_pre_hook(self._bwd_pre(u))
+            triggers = self._triggers(u)
+            for t in triggers:
+                t.register_forward_pre_hook(self._fwd_pre(u))
+                t.register_full_backward_pre_hook(self._bwd_pre(u))
+            triggers[-1].register_forward_hook(self._fwd_post(u))
             for p in u.params:
                 p.register_post_accumulate_grad_hook(self._grad_hook(u))
 
     def _fwd_pre(self, u: _Unit):
         def hook(module, args):
-            u.start_gather()
+            u.start_gather()  # no-op if already gathered
             u.finish_gather()
             # prefetch next unit's gather onto the fabric
             if self.prefetch and u.idx + 1 < len(self.units):
@@ -238,17 +250,15 @@ class Zero3Model(nn.Module):
 
     def _fwd_post(self, u: _Unit):
         def hook(module, args, output):
-            if not torch.is_grad_enabled():
-                u.free()
-            else:
-                u.free()  # re-gathered at backward
+            u.free()  # re-gathered at backward by _bwd_pre
         return hook
 
     def _bwd_pre(self, u: _Unit):
         def hook(module, grad_output):
             u.start_gather()
             u.finish_gather()
-            u.grad_pending = len(u.params)
+            if u.grad_pending == 0:
+                u.grad_pending = len(u.params)
             if self.prefetch and u.idx - 1 >= 0:
                 prv = self.units[u.idx - 1]
                 if prv.module is not self.model:

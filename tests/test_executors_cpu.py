@@ -1,0 +1,109 @@
+"""Pipeline and Spilled executors on CPU, plus the pipeline partitioner."""
+
+import pytest
+import torch
+import torch.nn as nn
+
+from saturn_amd import HParams, Strategy, Task
+from saturn_amd.executors.pipeline import PipelineExecutor
+from saturn_amd.executors.spilled import SpilledExecutor
+from saturn_amd.models import get_mlp_dataloader, get_mlp_model, mse_loss
+from saturn_amd.models.gptj import (
+    as_sequential,
+    get_gptj_model,
+    make_token_dataloader,
+    pretraining_loss,
+)
+from saturn_amd.parallel.pipeline import PipelinedModel, balance_by_params
+
+
+def test_balance_by_params():
+    seq = nn.Sequential(*[nn.Linear(16, 16) for _ in range(10)])
+    for n in (2, 3, 4):
+        bal = balance_by_params(seq, n)
+        assert sum(bal) == 10 and len(bal) == n
+        assert all(b >= 1 for b in bal)
+
+
+def test_pipelined_model_matches_unpipelined():
+    torch.manual_seed(0)
+    mk = lambda: get_gptj_model(
+        {"n_layer": 4, "n_embd": 64, "n_head": 2, "vocab_size": 128,
+         "n_ctx": 32, "rotary_dim": 8}
+    )
+    ref = mk()
+    m = mk()
+    seq = as_sequential(m)
+    pipe = PipelinedModel(seq, ["cpu", "cpu"], chunks=2)
+    x = torch.randint(0, 128, (4, 32))
+    l_ref = pretraining_loss(ref(x), x)
+    l_pipe = pretraining_loss(pipe(x), x)
+    assert abs(l_ref.item() - l_pipe.item()) < 1e-4
+    l_pipe.backward()
+    g = next(p.grad for p in pipe.parameters() if p.grad is not None)
+    assert torch.isfinite(g).all()
+
+
+def gptj_task(name, save_dir, batch_count=4):
+    return Task(
+        lambda kwargs=None: get_gptj_model(
+            {"n_layer": 4, "n_embd": 64, "n_head": 2, "vocab_size": 128,
+             "n_ctx": 32, "rotary_dim": 8}
+        ),
+        make_token_dataloader(batch_size=2, seq_len=32, vocab=128, n_batches=8),
+        pretraining_loss,
+        HParams(lr=1e-3, batch_count=batch_count),
+        name=name,
+        save_dir=save_dir,
+    )
+
+
+def test_pipeline_executor_search_and_execute(save_dir):
+    t = gptj_task("pipe_t", save_dir)
+    params, bt = PipelineExecutor.search(t, [0, 1], 920)
+    assert params is not None and "chunks" in params and bt > 0
+    t.strategies[2] = Strategy(PipelineExecutor, 2, params, bt * 4, batch_time=bt)
+    t.select_strategy(t.strategies[2])
+    PipelineExecutor.execute(t, [0, 1], 920, 2)
+    assert t.has_ckpt()
+
+
+def test_pipeline_rejects_single_gpu(save_dir):
+    t = gptj_task("pipe_1g", save_dir)
+    params, bt = PipelineExecutor.search(t, [0], 921)
+    assert params is None
+
+
+def test_spilled_executor_search_and_execute(save_dir):
+    t = gptj_task("spill_t", save_dir)
+    params, bt = SpilledExecutor.search(t, [0], 922)
+    assert params is not None and "partitions" in params
+    t.strategies[1] = Strategy(SpilledExecutor, 1, params, bt * 4, batch_time=bt)
+    t.select_strategy(t.strategies[1])
+    SpilledExecutor.execute(t, [0], 922, 2)
+    assert t.has_ckpt()
+
+
+def test_spilled_rejects_multi_gpu(save_dir):
+    t = gptj_task("spill_2g", save_dir)
+    params, bt = SpilledExecutor.search(t, [0, 1], 923)
+    assert params is None
+
+
+def test_llama_tiny_cpu_trains():
+    from saturn_amd.models.llama import get_llama_model, llama_loss
+
+    torch.manual_seed(0)
+    m = get_llama_model({"preset": "8b", "n_layer": 2, "n_ctx": 32,
+                         "vocab_size": 256})
+    opt = torch.optim.SGD(m.parameters(), lr=0.1)
+    x = torch.randint(0, 256, (2, 32))
+    l0 = None
+    for i in range(5):
+        loss = llama_loss(m(x), x)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        if i == 0:
+            l0 = loss.item()
+    assert loss.item() < l0
