@@ -1,0 +1,227 @@
+"""Makespan benchmark: run a BASELINE.json job batch end to end through the
+full stack (library -> trial search -> MILP solve -> interval orchestration)
+and report the batch makespan plus per-job throughput and the solver's plan.
+
+    python bench_makespan.py --config 2 --scale tiny --n-gpus 1
+
+Configs (BASELINE.json):
+  1  2-job MLP lr sweep            (CPU/gloo; the no-GPU plumbing config)
+  2  4-job GPT-2-small HPO sweep   (DDP-only library)
+  3  8-job Llama-3-8B lr/batch sweep  (DDP vs FSDP per job)
+  4  heterogeneous batch: GPT-2-XL + BERT-large + ViT-L + Llama-3-8B
+  5  2-job Llama-3-70B (pipeline + spilling)
+
+--scale tiny shrinks layer counts/batch quotas so a config finishes within
+minutes on one GPU; --scale full uses the named model sizes.  Rank-0 prints
+one JSON line with the makespan and the chosen plan.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import tempfile
+from timeit import default_timer as timer
+
+
+def log(*a):
+    print(*a, file=sys.stderr, flush=True)
+
+
+def build_tasks(config: int, scale: str, save_dir: str):
+    from saturn_amd import HParams, Task
+    from saturn_amd.models import (
+        get_bert_model,
+        get_gpt2_model,
+        get_gptj_model,
+        get_llama_model,
+        get_mlp_dataloader,
+        get_mlp_model,
+        gpt2_loss,
+        llama_loss,
+        make_image_dataloader,
+        make_mlm_dataloader,
+        make_token_dataloader,
+        mlm_loss,
+        mse_loss,
+        pretraining_loss,
+        vit_loss,
+        get_vit_model,
+    )
+
+    tiny = scale == "tiny"
+    bc = 8 if tiny else 64  # batches per job
+
+    def t(name, gm, gd, loss, lr, gpu_range, hints=None, kwargs=None):
+        return Task(
+            gm,
+            gd,
+            loss,
+            HParams(lr=lr, batch_count=bc, **(kwargs or {})),
+            gpu_range=gpu_range,
+            name=name,
+            hints=hints,
+            save_dir=save_dir,
+        )
+
+    if config == 1:
+        return [
+            t(f"mlp_lr{lr}", get_mlp_model, get_mlp_dataloader, mse_loss, lr, [1])
+            for lr in (1e-2, 1e-3)
+        ], ["ddp"]
+    if config == 2:
+        kw = {"preset": "small"}
+        if tiny:
+            kw["n_layer"] = 4
+        dl = make_token_dataloader(batch_size=8, seq_len=512 if not tiny else 256,
+                                   vocab=50257, n_batches=16)
+        return [
+            t(f"gpt2s_lr{i}", get_gpt2_model, dl, gpt2_loss, lr, None,
+              kwargs=dict(kw))
+            for i, lr in enumerate((1e-4, 3e-4, 1e-3, 3e-3))
+        ], ["ddp"]
+    if config == 3:
+        kw = {"preset": "8b"}
+        if tiny:
+            kw.update(n_layer=4, vocab_size=32000, n_ctx=512)
+        dl = make_token_dataloader(batch_size=4, seq_len=512,
+                                   vocab=kw.get("vocab_size", 128256),
+                                   n_batches=16)
+        return [
+            t(f"llama8b_{i}", get_llama_model, dl, llama_loss, lr, None,
+              kwargs=dict(kw))
+            for i, lr in enumerate(
+                (1e-5, 3e-5, 1e-4, 3e-4, 1e-3, 3e-3, 5e-4, 5e-5)
+            )
+        ], ["ddp", "fsdp"]
+    if config == 4:
+        seq = 256 if tiny else 512
+        nl = 4 if tiny else None
+        gpt2kw = {"preset": "xl"}
+        bertkw, vitkw, llamakw = {}, {}, {"preset": "8b"}
+        if tiny:
+            gpt2kw["n_layer"] = 4
+            bertkw["n_layer"] = 4
+            vitkw["n_layer"] = 4
+            llamakw.update(n_layer=4, vocab_size=32000, n_ctx=512)
+        return [
+            t("gpt2xl", get_gpt2_model,
+              make_token_dataloader(8, seq, 50257, 16), gpt2_loss, 3e-4,
+              None, kwargs=gpt2kw),
+            t("bert_large", get_bert_model,
+              make_mlm_dataloader(16, seq, 30522, 16), mlm_loss, 1e-4,
+              None, kwargs=bertkw),
+            t("vit_l", get_vit_model,
+              make_image_dataloader(16 if tiny else 32, 224, 16), vit_loss,
+              3e-4, None, kwargs=vitkw),
+            t("llama8b", get_llama_model,
+              make_token_dataloader(4, 512, llamakw.get("vocab_size", 128256), 16),
+              llama_loss, 1e-4, None, kwargs=llamakw),
+        ], ["ddp", "fsdp", "pipeline", "spilled"]
+    if config == 5:
+        kw = {"preset": "70b"}
+        if tiny:
+            kw.update(n_layer=8, vocab_size=32000, n_ctx=512)
+        dl = make_token_dataloader(2, 512, kw.get("vocab_size", 128256), 8)
+        return [
+            t(f"llama70b_{i}", get_llama_model, dl, llama_loss, lr, None,
+              kwargs=dict(kw))
+            for i, lr in enumerate((1e-5, 3e-5))
+        ], ["pipeline", "spilled", "fsdp"]
+    raise SystemExit(f"unknown config {config}")
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", type=int, default=1)
+    ap.add_argument("--scale", choices=["tiny", "full"], default="tiny")
+    ap.add_argument("--n-gpus", type=int, default=None)
+    ap.add_argument("--interval", type=float, default=None)
+    args = ap.parse_args()
+
+    import torch
+
+    n_gpus = args.n_gpus
+    if n_gpus is None:
+        n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 2
+
+    lib_dir = tempfile.mkdtemp(prefix="saturn_lib_")
+    save_dir = tempfile.mkdtemp(prefix="saturn_models_")
+    os.environ["SATURN_LIBRARY_PATH"] = lib_dir
+
+    from saturn_amd import register, search, solve, orchestrate
+    from saturn_amd.executors import (
+        DDPExecutor,
+        FSDPExecutor,
+        PipelineExecutor,
+        SpilledExecutor,
+    )
+    from saturn_amd.solver import apply_plan
+
+    by_name = {
+        "ddp": DDPExecutor,
+        "fsdp": FSDPExecutor,
+        "pipeline": PipelineExecutor,
+        "spilled": SpilledExecutor,
+    }
+    tasks, execs = build_tasks(args.config, args.scale, save_dir)
+    for nm in execs:
+        register(nm, by_name[nm])
+
+    log(f"[makespan] config {args.config} ({args.scale}): {len(tasks)} jobs, "
+        f"{len(execs)} techniques, {n_gpus} GPUs")
+    t0 = timer()
+    search(tasks, executor_names=execs, n_gpus=n_gpus, log_level=True)
+    t_search = timer() - t0
+
+    plan = solve(tasks, n_gpus=n_gpus, timeout=30)
+    apply_plan(tasks, plan)
+    plan_dump = [
+        {
+            "task": plan.task_names[i],
+            "executor": getattr(tasks[i].selected_strategy.executor, "name", None),
+            "gpus": plan.gpu_sets[i],
+            "start_s": round(plan.start_times[i], 2),
+            "est_runtime_s": round(plan.runtimes[i], 2),
+            "batch_time_s": tasks[i].selected_strategy.batch_time,
+        }
+        for i in range(len(tasks))
+    ]
+    log("[makespan] plan:", json.dumps(plan_dump, indent=1))
+
+    interval = args.interval
+    if interval is None:
+        interval = max(5.0, plan.makespan / 3)  # a few introspection cycles
+    t1 = timer()
+    orchestrate(tasks, interval=interval, n_gpus=n_gpus, solver_timeout=15)
+    makespan = timer() - t1
+
+    per_job = {
+        d["task"]: (None if d["batch_time_s"] is None
+                    else round(1.0 / d["batch_time_s"], 3))
+        for d in plan_dump
+    }
+    print(json.dumps({
+        "metric": "makespan_s",
+        "value": makespan,
+        "unit": "s",
+        "higher_is_better": False,
+        "n_gpus": n_gpus,
+        "config": {
+            "baseline_config": args.config,
+            "scale": args.scale,
+            "n_jobs": len(tasks),
+            "techniques": execs,
+        },
+        "search_time_s": round(t_search, 2),
+        "predicted_makespan_s": round(plan.makespan, 2),
+        "plan": plan_dump,
+        "batches_per_s_per_job": per_job,
+        "data": "synthetic",
+    }), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -26,10 +26,21 @@ def _to_sequential(task, model):
         return fn(model)
     if isinstance(model, nn.Sequential):
         return model
-    from saturn_amd.models.gptj import GPTJForCausalLM, as_sequential
+    from saturn_amd.models.gptj import GPTJForCausalLM
+    from saturn_amd.models.gptj import as_sequential as gptj_seq
 
     if isinstance(model, GPTJForCausalLM):
-        return as_sequential(model)
+        return gptj_seq(model)
+    from saturn_amd.models.gpt2 import GPT2ForCausalLM
+    from saturn_amd.models.gpt2 import as_sequential as gpt2_seq
+
+    if isinstance(model, GPT2ForCausalLM):
+        return gpt2_seq(model)
+    from saturn_amd.models.llama import LlamaForCausalLM
+    from saturn_amd.models.llama import as_sequential as llama_seq
+
+    if isinstance(model, LlamaForCausalLM):
+        return llama_seq(model)
     raise ValueError(
         "Pipeline executor needs an nn.Sequential model or a "
         "hints['to_sequential'] flattener."

@@ -1,6 +1,8 @@
 from .mlp import MLP, get_mlp_dataloader, get_mlp_model, mse_loss
 from .gpt2 import GPT2ForCausalLM, get_gpt2_model, gpt2_loss
 from .llama import LlamaForCausalLM, get_llama_model, llama_loss
+from .bert import BertForMaskedLM, get_bert_model, make_mlm_dataloader, mlm_loss
+from .vit import ViTForImageClassification, get_vit_model, make_image_dataloader, vit_loss
 from .gptj import (
     GPTJForCausalLM,
     get_gptj_model,
@@ -23,4 +25,12 @@ __all__ = [
     "LlamaForCausalLM",
     "get_llama_model",
     "llama_loss",
+    "BertForMaskedLM",
+    "get_bert_model",
+    "make_mlm_dataloader",
+    "mlm_loss",
+    "ViTForImageClassification",
+    "get_vit_model",
+    "make_image_dataloader",
+    "vit_loss",
 ]

@@ -104,6 +104,35 @@ class GPT2ForCausalLM(nn.Module):
         return self.lm_head(self.ln_f(x))
 
 
+class _GPT2Embed(nn.Module):
+    def __init__(self, wte, wpe):
+        super().__init__()
+        self.wte = wte
+        self.wpe = wpe
+
+    def forward(self, input_ids):
+        pos = torch.arange(input_ids.shape[1], device=input_ids.device)
+        return self.wte(input_ids) + self.wpe(pos)[None]
+
+
+class _GPT2Head(nn.Module):
+    def __init__(self, ln_f, lm_head):
+        super().__init__()
+        self.ln_f = ln_f
+        self.lm_head = lm_head
+
+    def forward(self, x):
+        return self.lm_head(self.ln_f(x))
+
+
+def as_sequential(model: "GPT2ForCausalLM") -> nn.Sequential:
+    """Flatten for the pipeline executor (shared parameters)."""
+    return nn.Sequential(
+        _GPT2Embed(model.wte, model.wpe), *model.h,
+        _GPT2Head(model.ln_f, model.lm_head)
+    )
+
+
 def gpt2_loss(logits, labels):
     return fused_cross_entropy(logits, labels, shift=True)
 

@@ -125,6 +125,32 @@ class LlamaForCausalLM(nn.Module):
         return self.lm_head(self.ln_f(x))
 
 
+class _LlamaEmbed(nn.Module):
+    def __init__(self, wte):
+        super().__init__()
+        self.wte = wte
+
+    def forward(self, input_ids):
+        return self.wte(input_ids)
+
+
+class _LlamaHead(nn.Module):
+    def __init__(self, ln_f, lm_head):
+        super().__init__()
+        self.ln_f = ln_f
+        self.lm_head = lm_head
+
+    def forward(self, x):
+        return self.lm_head(self.ln_f(x))
+
+
+def as_sequential(model: "LlamaForCausalLM") -> nn.Sequential:
+    """Flatten for the pipeline executor (shared parameters)."""
+    return nn.Sequential(
+        _LlamaEmbed(model.wte), *model.h, _LlamaHead(model.ln_f, model.lm_head)
+    )
+
+
 def llama_loss(logits, labels):
     return fused_cross_entropy(logits, labels, shift=True)
 

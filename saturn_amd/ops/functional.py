@@ -236,15 +236,24 @@ def attention_math(q, k, v, causal: bool = True) -> torch.Tensor:
     return torch.matmul(p, v.float()).to(q.dtype)
 
 
-def causal_attention(q, k, v) -> torch.Tensor:
-    """Dispatch: fused CDNA4 flash kernel on GPU (when built), math path on
-    CPU.  Handles GQA (k/v with fewer heads) by expansion."""
+def _attention(q, k, v, causal: bool) -> torch.Tensor:
     if q.is_cuda:
         from saturn_amd.ops import flash  # local import: optional kernel
 
-        return flash.flash_attention(q, k, v, causal=True)
+        return flash.flash_attention(q, k, v, causal=causal)
     if k.shape[1] != q.shape[1]:
         rep = q.shape[1] // k.shape[1]
         k = k.repeat_interleave(rep, dim=1)
         v = v.repeat_interleave(rep, dim=1)
-    return attention_math(q, k, v, causal=True)
+    return attention_math(q, k, v, causal=causal)
+
+
+def causal_attention(q, k, v) -> torch.Tensor:
+    """Dispatch: fused CDNA4 flash kernel on GPU (when built), math path on
+    CPU.  Handles GQA (k/v with fewer heads) by expansion."""
+    return _attention(q, k, v, causal=True)
+
+
+def full_attention(q, k, v) -> torch.Tensor:
+    """Bidirectional (encoder) attention — BERT/ViT."""
+    return _attention(q, k, v, causal=False)
