@@ -75,7 +75,7 @@ class GPTJAttention(nn.Module):
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
         o = causal_attention(q, k, v)
-        o = o.transpose(1, 2).reshape(B, T, E)
+        o = o.transpose(1, 2).reshape(B, T, -1)  # -1: heads may be TP-sharded
         return self.out_proj(o)
 
 
