@@ -14,6 +14,7 @@ optimizer step is the fused multi-tensor HIP kernel when running on GPU
 
 from __future__ import annotations
 
+import os
 from timeit import default_timer as timer
 from typing import Any, Dict, List, Optional, Tuple
 
@@ -65,7 +66,12 @@ def _ddp_worker(
         model = model.to(device=device, dtype=dtype)
         model.train()
         bucket_mb = float((params or {}).get("bucket_mb", 64.0))
-        ddp = BucketedDDP(model, bucket_mb=bucket_mb)
+        comm = None
+        if backend == "nccl" and os.environ.get("SATURN_NATIVE_COMM", "0") == "1":
+            from saturn_amd.comm import create_comm
+
+            comm = create_comm(rank, world)
+        ddp = BucketedDDP(model, bucket_mb=bucket_mb, comm=comm)
         optimizer = _make_optimizer(task, model)
 
         ckpt = task.load_checkpoint()

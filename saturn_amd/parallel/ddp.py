@@ -61,18 +61,33 @@ class BucketedDDP(torch.nn.Module):
         bucket_mb: float = 64.0,
         process_group=None,
         grad_dtype: Optional[torch.dtype] = None,
+        comm=None,
     ) -> None:
+        """``comm``: an optional saturn_amd native RcclComm; when given,
+        bucket all-reduces run on its dedicated HIP stream (ncclAvg)
+        instead of torch.distributed's process group."""
         super().__init__()
         self.module = module
         self.pg = process_group
-        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self.comm = comm
+        if comm is not None:
+            self.world = comm.world
+        else:
+            self.world = (
+                dist.get_world_size(process_group) if dist.is_initialized() else 1
+            )
 
         # Broadcast initial parameters from rank 0 (reference relies on the
         # DDP ctor for this, DDP.py:90).
         if self.world > 1:
             with torch.no_grad():
-                for p in module.parameters():
-                    dist.broadcast(p.data, src=0, group=self.pg)
+                if comm is not None:
+                    for p in module.parameters():
+                        comm.broadcast(p.data, 0)
+                    comm.join()
+                else:
+                    for p in module.parameters():
+                        dist.broadcast(p.data, src=0, group=self.pg)
 
         params = [p for p in module.parameters() if p.requires_grad]
         bucket_bytes = int(bucket_mb * 1024 * 1024)
@@ -110,13 +125,24 @@ class BucketedDDP(torch.nn.Module):
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0 and self.world > 1:
-            b.work = dist.all_reduce(b.flat, async_op=True, group=self.pg)
+            if self.comm is not None:
+                # enqueued on the native engine's comm stream, fenced
+                # against the producing compute stream — overlaps backward
+                self.comm.all_reduce(b.flat, True)
+            else:
+                b.work = dist.all_reduce(b.flat, async_op=True, group=self.pg)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
 
     def grad_sync(self) -> None:
         """Wait for in-flight bucket all-reduces and average."""
+        if self.comm is not None:
+            if self.world > 1:
+                self.comm.join()  # compute stream waits the comm stream
+            for b in self.buckets:
+                b.reset()
+            return
         inv = 1.0 / self.world
         for b in self.buckets:
             if b.work is not None:

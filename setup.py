@@ -32,6 +32,16 @@ ext = CUDAExtension(
     },
 )
 
+comm_ext = CUDAExtension(
+    name="saturn_amd._comm",
+    sources=[os.path.join("saturn_amd", "comm", "csrc", "rccl_comm.cpp")],
+    libraries=["rccl"],
+    extra_compile_args={
+        "cxx": ["-O3", "-std=c++17"],
+        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+    },
+)
+
 setup(
     name="saturn_amd",
     version="0.1.0",
@@ -45,9 +55,10 @@ setup(
         "saturn_amd.executors",
         "saturn_amd.parallel",
         "saturn_amd.ops",
+        "saturn_amd.comm",
         "saturn_amd.models",
         "saturn_amd.utils",
     ],
-    ext_modules=[ext],
+    ext_modules=[ext, comm_ext],
     cmdclass={"build_ext": BuildExtension},
 )
