@@ -33,7 +33,7 @@ def main() -> None:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--batch", type=int, default=8, help="per-GPU batch size")
+    ap.add_argument("--batch", type=int, default=16, help="per-GPU batch size")
     ap.add_argument("--seq", type=int, default=512)
     ap.add_argument("--layers", type=int, default=28)
     ap.add_argument("--model", type=str, default="gptj-6b")

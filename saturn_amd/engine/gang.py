@@ -62,7 +62,7 @@ def call_in_subprocess(
     ``env`` merged into its environment; return its result or re-raise the
     child error here with the original traceback."""
     out_q = _CTX.Queue()
-    payload = dill.dumps((fn, args, kwargs))
+    payload = dill.dumps((fn, args, kwargs), recurse=True)
     p = _CTX.Process(
         target=_child_entry, args=(payload, env or {}, out_q), daemon=False
     )

@@ -51,7 +51,7 @@ def gang_spawn(
     from saturn_amd.utils.ports import rendezvous_env
 
     q = _CTX.Queue()
-    payload = dill.dumps((fn, args, kwargs))
+    payload = dill.dumps((fn, args, kwargs), recurse=True)
     procs: List[mp.Process] = []
     for rank in range(world_size):
         env = rendezvous_env(tid, rank, world_size)

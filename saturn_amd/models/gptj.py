@@ -27,6 +27,7 @@ from saturn_amd.ops.functional import (
     FusedLayerNorm,
     apply_rope,
     causal_attention,
+    fused_add3,
     fused_cross_entropy,
     rope_tables,
 )
@@ -101,7 +102,7 @@ class GPTJBlock(nn.Module):
 
     def forward(self, x):
         h = self.ln_1(x)
-        return x + self.attn(h) + self.mlp(h)
+        return fused_add3(x, self.attn(h), self.mlp(h))
 
 
 class GPTJForCausalLM(nn.Module):

@@ -28,6 +28,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor o, at::Tensor lse,
                                  bool causal);
+at::Tensor add3(at::Tensor a, at::Tensor b, at::Tensor c);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -41,4 +42,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_apply", &samd::rope_apply, "fused rotary embedding (in-place)");
   m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
   m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward");
+  m.def("add3", &samd::add3, "fused 3-way residual add");
 }

@@ -257,3 +257,24 @@ def causal_attention(q, k, v) -> torch.Tensor:
 def full_attention(q, k, v) -> torch.Tensor:
     """Bidirectional (encoder) attention — BERT/ViT."""
     return _attention(q, k, v, causal=False)
+
+
+# ---------------------------------------------------------------------------
+# Fused 3-way residual add (K7)
+# ---------------------------------------------------------------------------
+class _Add3Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b, c):
+        ext = require_ext()
+        return ext.add3(a.contiguous(), b.contiguous(), c.contiguous())
+
+    @staticmethod
+    def backward(ctx, g):
+        return g, g, g
+
+
+def fused_add3(a, b, c):
+    """out = a + b + c in one HBM pass (GPT-J parallel block residual)."""
+    if a.is_cuda:
+        return _Add3Fn.apply(a, b, c)
+    return a + b + c

@@ -25,6 +25,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "cross_entropy.hip"),
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "attention.hip"),
+        os.path.join(CSRC, "add3.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

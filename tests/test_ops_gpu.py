@@ -278,3 +278,15 @@ def test_flash_attention_bwd_vs_fp32():
     assert rel_err(q.grad, q32.grad) < 5e-2
     assert rel_err(k.grad, k32.grad) < 5e-2
     assert rel_err(v.grad, v32.grad) < 5e-2
+
+
+def test_add3_matches_reference():
+    ext = requires_ext()
+    torch.manual_seed(0)
+    for n in (8 * 1000, 12345):  # vector path + ragged tail
+        a = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        b = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        c = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+        out = ext.add3(a, b, c)
+        ref = (a.float() + b.float() + c.float()).to(torch.bfloat16)
+        assert (out.float() - ref.float()).abs().max().item() < 1e-1
