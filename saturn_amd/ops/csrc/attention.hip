@@ -33,18 +33,19 @@ constexpr int QBLK = 64;   // q rows per block
 constexpr int KBLK = 32;   // keys per kv tile
 constexpr int WROWS = 16;  // q rows per wave
 
-// LDS layout (per block), bf16 K/V tiles + per-wave fp32 scratch:
-//   K[32][D], V[32][D] (+8 bf16 row pad to break bank conflicts)
-//   per wave: P[16][32] bf16, m[16], l[16], alpha[16] fp32
+// LDS layout (per block), bf16 K/V tiles + per-wave P relayout buffer:
+//   K[32][D], V[32][D] (+8 bf16 row pad: row stride = 16 B mod 256 B, so
+//   the 16-lane b128 groups of the K-fragment reads land on distinct slots)
+//   per wave: P[16][32] bf16 (score D-layout -> A-fragment relayout).
+// Softmax state (m, l) lives in registers: row r of a wave's 16 q rows is
+// owned by quadrant r>>2 at accumulator reg r&3; its 16 score columns live
+// in that quadrant's 16 lanes, so row max/sum reduce with 4 shfl_xor steps
+// and never touch LDS.
 template <int D>
 struct AttnLds {
   unsigned short k[KBLK][D + 8];
   unsigned short v[KBLK][D + 8];
   unsigned short p[4][WROWS][KBLK];
-  float s[4][WROWS][KBLK + 2];
-  float m[4][WROWS];
-  float l[4][WROWS];
-  float alpha[4][WROWS];
 };
 
 template <int D>
@@ -83,11 +84,9 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
   for (int dt = 0; dt < D / 16; ++dt) otile[dt] = {0.f, 0.f, 0.f, 0.f};
 
-  // ---- init softmax state
-  if (lane < WROWS) {
-    lds.m[wid][lane] = -INFINITY;
-    lds.l[wid][lane] = 0.f;
-  }
+  // ---- softmax state in registers (per quadrant: rows qg*4+reg)
+  float m_reg[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+  float l_reg[4] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? (q0_block + QBLK) : T;
   for (int kv0 = 0; kv0 < kv_end; kv0 += KBLK) {
@@ -118,52 +117,36 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_q[ds], bk0, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_q[ds], bk1, acc1, 0, 0, 0);
     }
-    // write S to wave-private LDS with causal mask
+    // ---- online softmax fully in registers.  Row (qg*4+reg)'s 32
+    // scores sit in acc0[reg]/acc1[reg] across the quadrant's 16 lanes;
+    // reduce with 4 shfl_xor steps (no LDS round trip).
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
-      int row = qg * 4 + reg;           // S row (q row = q0 + row)
-      int qrow = q0 + row;
+      const int row = qg * 4 + reg;
+      const int qrow = q0 + row;
       float s0 = acc0[reg] * scale;
       float s1 = acc1[reg] * scale;
       if (causal) {
         if (kv0 + r > qrow) s0 = -INFINITY;
         if (kv0 + 16 + r > qrow) s1 = -INFINITY;
       }
-      lds.s[wid][row][r] = s0;
-      lds.s[wid][row][16 + r] = s1;
-    }
-
-    // ---- online softmax (wave-private; lanes l, l+16, l+32, l+48 share row r)
-    {
-      float tm = -INFINITY;
+      float tm = fmaxf(s0, s1);
 #pragma unroll
-      for (int c = 0; c < 8; ++c)
-        tm = fmaxf(tm, lds.s[wid][r][qg * 8 + c]);
-      tm = fmaxf(tm, __shfl_xor(tm, 16, 64));
-      tm = fmaxf(tm, __shfl_xor(tm, 32, 64));
-      float m_old = lds.m[wid][r];
-      float m_new = fmaxf(m_old, tm);
-      float al = (m_old == -INFINITY) ? 0.f : __expf(m_old - m_new);
-      float ps = 0.f;
+      for (int off = 1; off < 16; off <<= 1)
+        tm = fmaxf(tm, __shfl_xor(tm, off, 64));
+      const float m_new = fmaxf(m_reg[reg], tm);
+      const float al =
+          (m_reg[reg] == -INFINITY) ? 0.f : __expf(m_reg[reg] - m_new);
+      const float p0 = (s0 == -INFINITY) ? 0.f : __expf(s0 - m_new);
+      const float p1 = (s1 == -INFINITY) ? 0.f : __expf(s1 - m_new);
+      float ps = p0 + p1;
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        float sv = lds.s[wid][r][qg * 8 + c];
-        float pv = (sv == -INFINITY) ? 0.f : __expf(sv - m_new);
-        lds.p[wid][r][qg * 8 + c] = f2us(pv);
-        ps += pv;
-      }
-      ps += __shfl_xor(ps, 16, 64);
-      ps += __shfl_xor(ps, 32, 64);
-      if (lane < WROWS) {
-        lds.m[wid][r] = m_new;
-        lds.l[wid][r] = lds.l[wid][r] * al + ps;
-        lds.alpha[wid][r] = al;
-      }
-    }
-    // rescale O accumulators by this row's alpha
-#pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      float al = lds.alpha[wid][qg * 4 + reg];
+      for (int off = 1; off < 16; off <<= 1)
+        ps += __shfl_xor(ps, off, 64);
+      l_reg[reg] = l_reg[reg] * al + ps;
+      m_reg[reg] = m_new;
+      lds.p[wid][row][r] = f2us(p0);
+      lds.p[wid][row][16 + r] = f2us(p1);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) otile[dt][reg] *= al;
     }
@@ -185,14 +168,16 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     int row = qg * 4 + reg;
-    float linv = 1.f / lds.l[wid][row];
+    float linv = 1.f / l_reg[reg];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt)
       Oh[(long)(q0 + row) * D + dt * 16 + r] = f2us(otile[dt][reg] * linv);
   }
-  if (lane < WROWS && LSE != nullptr) {
-    LSE[bh * (long)T + q0 + lane] =
-        lds.m[wid][lane] + __logf(lds.l[wid][lane]);
+  if (r == 0 && LSE != nullptr) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg)
+      LSE[bh * (long)T + q0 + qg * 4 + reg] =
+          m_reg[reg] + __logf(l_reg[reg]);
   }
 }
 
