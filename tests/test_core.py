@@ -138,3 +138,18 @@ def test_public_api_surface():
                 "register", "deregister", "retrieve", "search", "orchestrate",
                 "solve", "Plan"]:
         assert hasattr(saturn_amd, sym)
+
+
+def test_text_dataloader_cache_roundtrip(tmp_path):
+    from saturn_amd.models.data import load_text_dataset, make_text_dataloader
+
+    p = tmp_path / "corpus.txt"
+    p.write_text("hello world, this is a tiny corpus for windowing. " * 50)
+    w1 = load_text_dataset(str(p), context_length=64)
+    assert w1.shape[1] == 64 and w1.shape[0] > 10
+    assert (tmp_path / "corpus.txt.ctx64.npz").exists()
+    w2 = load_text_dataset(str(p), context_length=64)  # cache hit
+    assert torch.equal(w1, w2)
+    dl = make_text_dataloader(str(p), batch_size=4, context_length=64)()
+    x, y = next(iter(dl))
+    assert x.shape == (4, 64) and torch.equal(x, y)
