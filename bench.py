@@ -70,7 +70,11 @@ def main() -> None:
     cfg = GPTJConfig(n_layer=args.layers, n_ctx=args.seq)
     torch.manual_seed(1234)
     log(f"[rank {rank}] building GPT-J ({cfg.n_layer} layers) on {device}...")
-    model = GPTJForCausalLM(cfg).to(device=device, dtype=dtype)
+    # build directly on the device: at world 8 a host-side build would
+    # transiently hold 8 x 24 GB fp32 replicas in DRAM
+    with device:
+        model = GPTJForCausalLM(cfg)
+    model = model.to(dtype=dtype)
     model.train()
     n_params = sum(p.numel() for p in model.parameters())
     ddp = BucketedDDP(model, bucket_mb=args.bucket_mb)
