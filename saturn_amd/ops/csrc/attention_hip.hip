@@ -62,9 +62,12 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int q0_block = blockIdx.x * QBLK;        // block's first q row
+  // grid = (B*H, T/QBLK): bh varies fastest so the causal q-block work
+  // imbalance spreads across XCDs (dispatcher places block b on XCD b%8;
+  // with q-block fastest, XCD k would get ONLY q-block-index k's length)
+  const int q0_block = blockIdx.y * QBLK;        // block's first q row
   const int q0 = q0_block + wid * WROWS;         // wave's first q row
-  const long bh = blockIdx.y;
+  const long bh = blockIdx.x;
   const unsigned short* Qh = Q + bh * (long)T * D;
   const unsigned short* Kh = K + bh * (long)T * D;
   const unsigned short* Vh = V + bh * (long)T * D;
@@ -201,7 +204,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
-  dim3 grid(T / QBLK, B * H), block(256);
+  dim3 grid(B * H, T / QBLK), block(256);
 
 #define LAUNCH(DD)                                                          \
   do {                                                                      \
@@ -272,8 +275,9 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   const int wid = threadIdx.x >> 6;
   const int r = lane & 15;
   const int qg = lane >> 4;
-  const int kv0 = blockIdx.x * KB;
-  const long bh = blockIdx.y;
+  // bh fastest (see forward): spreads the causal kv-block imbalance
+  const int kv0 = blockIdx.y * KB;
+  const long bh = blockIdx.x;
   const unsigned short* Qh = Q + bh * (long)T * D;
   const unsigned short* Kh = K + bh * (long)T * D;
   const unsigned short* Vh = V + bh * (long)T * D;
@@ -440,7 +444,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   auto dv = at::empty_like(v);
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
-  dim3 grid(T / KB, B * H), block(256);
+  dim3 grid(B * H, T / KB), block(256);
 
 #define LAUNCH_B(DD)                                                         \
   do {                                                                       \
