@@ -94,6 +94,8 @@ def _ddp_worker(
             task.get_iterator() if not trial else task.get_fresh_iterator()
         )
 
+        last_loss = [0.0]
+
         def step(batch) -> None:
             x, y = batch
             x = x.to(device, non_blocking=True)
@@ -104,6 +106,7 @@ def _ddp_worker(
             ddp.grad_sync()
             optimizer.step()
             ddp.zero_grad_buffers()
+            last_loss[0] = loss.detach()
 
         def next_batch():
             nonlocal it
@@ -138,6 +141,12 @@ def _ddp_worker(
             if device.type == "cuda":
                 torch.cuda.synchronize()
             if rank == 0:
+                import logging
+
+                logging.getLogger(__name__).info(
+                    "task %s: %d batches done, loss %.4f",
+                    task.name, batch_count, float(last_loss[0]),
+                )
                 task.save_checkpoint(model, optimizer)
             import torch.distributed as dist
 
