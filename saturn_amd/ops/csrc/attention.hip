@@ -242,10 +242,12 @@ constexpr int KB = 64;   // keys per block
 constexpr int QT = 32;   // q rows per tile
 }
 
+// V never enters LDS: its backward uses are wave-private A-fragments
+// (16 own rows, 16-B vector reads) that L2 serves directly — dropping the
+// tile keeps the D=256 LDS footprint at ~78 KB so TWO blocks fit per CU.
 template <int D>
 struct AttnBwdLds {
   unsigned short kt[bwd::KB][D + 8];
-  unsigned short vt[bwd::KB][D + 8];
   unsigned short qt[bwd::QT][D + 8];
   unsigned short dot[bwd::QT][D + 8];
   unsigned short pt[4][16][bwd::QT + 8];   // per-wave P^T tile
@@ -255,7 +257,7 @@ struct AttnBwdLds {
 };
 
 template <int D>
-__launch_bounds__(256, 1)
+__launch_bounds__(256, 2)
 __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 const unsigned short* __restrict__ Q,
                                 const unsigned short* __restrict__ K,
@@ -284,7 +286,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
 
-  // ---- stage K/V tiles once (they live for the whole block)
+  // ---- stage the K tile once (lives for the whole block; V stays in L2)
   {
     constexpr int CHUNKS = (KB * D) / (256 * 8);
 #pragma unroll
@@ -293,8 +295,6 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       int row = idx / D, col = idx % D;
       *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
           *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
-      *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
-          *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
     }
   }
 
@@ -337,7 +337,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         bf16x8 ak = *reinterpret_cast<const bf16x8*>(
             &lds.kt[wid * 16 + r][ds * 32 + qg * 8]);
         bf16x8 av = *reinterpret_cast<const bf16x8*>(
-            &lds.vt[wid * 16 + r][ds * 32 + qg * 8]);
+            Vh + (long)(kv0 + wid * 16 + r) * D + ds * 32 + qg * 8);
         bf16x8 bq = *reinterpret_cast<const bf16x8*>(
             &lds.qt[n * 16 + r][ds * 32 + qg * 8]);
         bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
