@@ -243,12 +243,15 @@ constexpr int KB = 64;   // keys per block
 constexpr int QT = 32;   // q rows per tile
 }
 
-// V never enters LDS: its backward uses are wave-private A-fragments
-// (16 own rows, 16-B vector reads) that L2 serves directly — dropping the
-// tile keeps the D=256 LDS footprint at ~78 KB so TWO blocks fit per CU.
+// V placement is D-dependent: at D=256 the full struct would be ~112 KB
+// (one block/CU), so V stays in L2 and its wave-private A-fragments read
+// global 16-B vectors — measured +39% at the GPT-J backward shape.  At
+// D<=128 the struct fits two blocks/CU WITH the V tile, and the LDS copy
+// is faster (keeping it avoided a measured regression at T=2048/D=128).
 template <int D>
 struct AttnBwdLds {
   unsigned short kt[bwd::KB][D + 8];
+  unsigned short vt[D <= 128 ? bwd::KB : 1][D + 8];
   unsigned short qt[bwd::QT][D + 8];
   unsigned short dot[bwd::QT][D + 8];
   unsigned short pt[4][16][bwd::QT + 8];   // per-wave P^T tile
@@ -287,7 +290,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
 
-  // ---- stage the K tile once (lives for the whole block; V stays in L2)
+  // ---- stage the K tile once (and V when it fits; see struct comment)
   {
     constexpr int CHUNKS = (KB * D) / (256 * 8);
 #pragma unroll
@@ -296,6 +299,9 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       int row = idx / D, col = idx % D;
       *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
           *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
+      if constexpr (D <= 128)
+        *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
+            *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
     }
   }
 
@@ -337,8 +343,13 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       for (int ds = 0; ds < D / 32; ++ds) {
         bf16x8 ak = *reinterpret_cast<const bf16x8*>(
             &lds.kt[wid * 16 + r][ds * 32 + qg * 8]);
-        bf16x8 av = *reinterpret_cast<const bf16x8*>(
-            Vh + (long)(kv0 + wid * 16 + r) * D + ds * 32 + qg * 8);
+        bf16x8 av;
+        if constexpr (D <= 128)
+          av = *reinterpret_cast<const bf16x8*>(
+              &lds.vt[wid * 16 + r][ds * 32 + qg * 8]);
+        else
+          av = *reinterpret_cast<const bf16x8*>(
+              Vh + (long)(kv0 + wid * 16 + r) * D + ds * 32 + qg * 8);
         bf16x8 bq = *reinterpret_cast<const bf16x8*>(
             &lds.qt[n * 16 + r][ds * 32 + qg * 8]);
         bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
