@@ -33,6 +33,18 @@ constexpr int QBLK = 64;   // q rows per block
 constexpr int KBLK = 32;   // keys per kv tile
 constexpr int WROWS = 16;  // q rows per wave
 
+// Element strides of a [B, H, T, D] tensor with stride(D) == 1.  Kernels
+// address through these, so transposed views ([B,T,H,D] physical — the
+// model's natural projection layout) need NO .contiguous() copies, and the
+// output is written straight into [B,T,H,D] physical so the model's
+// transpose+reshape after attention is a free view.
+struct TStr {
+  long sb, sh, st;
+};
+static TStr str_of(const at::Tensor& t) {
+  return TStr{t.stride(0), t.stride(1), t.stride(2)};
+}
+
 // LDS layout (per block), bf16 K/V tiles + per-wave P relayout buffer:
 //   K[32][D], V[32][D] (+8 bf16 row pad: row stride = 16 B mod 256 B, so
 //   the 16-lane b128 groups of the K-fragment reads land on distinct slots)
@@ -55,7 +67,8 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
                                 const unsigned short* __restrict__ V,
                                 unsigned short* __restrict__ O,
                                 float* __restrict__ LSE, int T, int n_heads,
-                                float scale, int causal) {
+                                float scale, int causal, TStr qs, TStr ks,
+                                TStr vs, TStr os) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnLds<D>& lds = *reinterpret_cast<AttnLds<D>*>(smem);
 
@@ -67,9 +80,10 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int q0_block = blockIdx.y * QBLK;        // block's first q row
   const int q0 = q0_block + wid * WROWS;         // wave's first q row
   const long bh = blockIdx.x;
-  const unsigned short* Qh = Q + bh * (long)T * D;
-  const unsigned short* Kh = K + bh * (long)T * D;
-  const unsigned short* Vh = V + bh * (long)T * D;
+  const long b = bh / n_heads, h = bh % n_heads;
+  const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
 
   const int r = lane & 15;       // fragment row/col index
   const int qg = lane >> 4;      // quadrant 0..3
@@ -78,7 +92,7 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   bf16x8 frag_q[D / 32];
 #pragma unroll
   for (int ds = 0; ds < D / 32; ++ds) {
-    const unsigned short* src = Qh + (long)(q0 + r) * D + ds * 32 + qg * 8;
+    const unsigned short* src = Qh + (long)(q0 + r) * qs.st + ds * 32 + qg * 8;
     frag_q[ds] = *reinterpret_cast<const bf16x8*>(src);
   }
 
@@ -102,9 +116,11 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
         int idx = (c * 256 + threadIdx.x) * 8;
         int row = idx / D, col = idx % D;
         *reinterpret_cast<bf16x8*>(&lds.k[row][col]) =
-            *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
+            *reinterpret_cast<const bf16x8*>(
+                Kh + (long)(kv0 + row) * ks.st + col);
         *reinterpret_cast<bf16x8*>(&lds.v[row][col]) =
-            *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
+            *reinterpret_cast<const bf16x8*>(
+                Vh + (long)(kv0 + row) * vs.st + col);
       }
     }
     __syncthreads();
@@ -167,14 +183,15 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   }
 
   // ---- epilogue: O /= l, write bf16; LSE = m + log(l)
-  unsigned short* Oh = O + bh * (long)T * D;
+  unsigned short* Oh = O + b * os.sb + h * os.sh;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     int row = qg * 4 + reg;
     float linv = 1.f / l_reg[reg];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt)
-      Oh[(long)(q0 + row) * D + dt * 16 + r] = f2us(otile[dt][reg] * linv);
+      Oh[(long)(q0 + row) * os.st + dt * 16 + r] =
+          f2us(otile[dt][reg] * linv);
   }
   if (r == 0 && LSE != nullptr) {
 #pragma unroll
@@ -192,14 +209,21 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
               "attn_fwd: bf16 only");
   TORCH_CHECK(q.dim() == 4, "attn_fwd: [B, H, T, D]");
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "attn_fwd: innermost dim must be dense");
   const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
             D = (int)q.size(3);
   TORCH_CHECK(k.size(2) == T, "attn_fwd: q/k length mismatch");
   TORCH_CHECK(T % QBLK == 0, "attn_fwd: T must be a multiple of 64");
   TORCH_CHECK(D == 64 || D == 128 || D == 256, "attn_fwd: D in {64,128,256}");
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(q.data_ptr()) & 15) == 0 &&
+              q.stride(2) % 8 == 0 && k.stride(2) % 8 == 0 &&
+              v.stride(2) % 8 == 0, "attn_fwd: rows must be 16-B aligned");
 
-  auto o = at::empty_like(q);
+  // output physically [B, T, H, D]: the model's post-attention
+  // transpose(1,2).reshape is then a zero-copy view
+  auto o_phys = at::empty({B, T, H, D}, q.options());
+  auto o = o_phys.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
@@ -214,7 +238,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                        reinterpret_cast<const unsigned short*>(k.data_ptr()), \
                        reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                        reinterpret_cast<unsigned short*>(o.data_ptr()),     \
-                       lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0); \
+                       lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0,  \
+                       str_of(q), str_of(k), str_of(v), str_of(o));         \
   } while (0)
 
   if (D == 64) LAUNCH(64);
@@ -270,7 +295,9 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 float* __restrict__ dQ,
                                 unsigned short* __restrict__ dK,
                                 unsigned short* __restrict__ dV, int T,
-                                float scale, int causal) {
+                                int n_heads, float scale, int causal,
+                                TStr dos, TStr qs, TStr ks, TStr vs,
+                                TStr dqs, TStr dks, TStr dvs) {
   using namespace bwd;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnBwdLds<D>& lds = *reinterpret_cast<AttnBwdLds<D>*>(smem);
@@ -282,9 +309,10 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   // bh fastest (see forward): spreads the causal kv-block imbalance
   const int kv0 = blockIdx.y * KB;
   const long bh = blockIdx.x;
-  const unsigned short* Qh = Q + bh * (long)T * D;
-  const unsigned short* Kh = K + bh * (long)T * D;
-  const unsigned short* Vh = V + bh * (long)T * D;
+  const long b = bh / n_heads, h = bh % n_heads;
+  const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
   const unsigned short* dOh = dO + bh * (long)T * D;
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
@@ -297,10 +325,12 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       int idx = (c * 256 + threadIdx.x) * 8;
       int row = idx / D, col = idx % D;
       *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
-          *reinterpret_cast<const bf16x8*>(Kh + (long)(kv0 + row) * D + col);
+          *reinterpret_cast<const bf16x8*>(
+              Kh + (long)(kv0 + row) * ks.st + col);
       if constexpr (D <= 128)
         *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
-            *reinterpret_cast<const bf16x8*>(Vh + (long)(kv0 + row) * D + col);
+            *reinterpret_cast<const bf16x8*>(
+                Vh + (long)(kv0 + row) * vs.st + col);
     }
   }
 
@@ -322,9 +352,11 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         int idx = (c * 256 + threadIdx.x) * 8;
         int row = idx / D, col = idx % D;
         *reinterpret_cast<bf16x8*>(&lds.qt[row][col]) =
-            *reinterpret_cast<const bf16x8*>(Qh + (long)(q0 + row) * D + col);
+            *reinterpret_cast<const bf16x8*>(
+                Qh + (long)(q0 + row) * qs.st + col);
         *reinterpret_cast<bf16x8*>(&lds.dot[row][col]) =
-            *reinterpret_cast<const bf16x8*>(dOh + (long)(q0 + row) * D + col);
+            *reinterpret_cast<const bf16x8*>(
+                dOh + (long)(q0 + row) * dos.st + col);
       }
       if (threadIdx.x < QT) {
         lds.lse_t[threadIdx.x] = lse_h[q0 + threadIdx.x];
@@ -348,7 +380,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
               &lds.vt[wid * 16 + r][ds * 32 + qg * 8]);
         else
           av = *reinterpret_cast<const bf16x8*>(
-              Vh + (long)(kv0 + wid * 16 + r) * D + ds * 32 + qg * 8);
+              Vh + (long)(kv0 + wid * 16 + r) * vs.st + ds * 32 + qg * 8);
         bf16x8 bq = *reinterpret_cast<const bf16x8*>(
             &lds.qt[n * 16 + r][ds * 32 + qg * 8]);
         bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
@@ -415,7 +447,8 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
 #pragma unroll
         for (int reg = 0; reg < 4; ++reg) {
           const int q_glob = q0 + n * 16 + qg * 4 + reg;
-          atomicAdd(&dQ[(bh * (long)T + q_glob) * D + dt * 16 + r],
+          atomicAdd(&dQ[b * dqs.sb + h * dqs.sh + (long)q_glob * dqs.st +
+                        dt * 16 + r],
                     acc_dq[reg]);
         }
       }
@@ -423,15 +456,15 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   }
 
   // ---- epilogue: write dK, dV (bf16)
-  unsigned short* dKh = dK + bh * (long)T * D;
-  unsigned short* dVh = dV + bh * (long)T * D;
+  unsigned short* dKh = dK + b * dks.sb + h * dks.sh;
+  unsigned short* dVh = dV + b * dvs.sb + h * dvs.sh;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     const int key_glob = kv0 + wid * 16 + qg * 4 + reg;
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
-      dKh[(long)key_glob * D + dt * 16 + r] = f2us(acc_dk[dt][reg]);
-      dVh[(long)key_glob * D + dt * 16 + r] = f2us(acc_dv[dt][reg]);
+      dKh[(long)key_glob * dks.st + dt * 16 + r] = f2us(acc_dk[dt][reg]);
+      dVh[(long)key_glob * dvs.st + dt * 16 + r] = f2us(acc_dv[dt][reg]);
     }
   }
 }
@@ -445,13 +478,18 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
             D = (int)q.size(3);
   TORCH_CHECK(T % KB == 0, "attn_bwd: T must be a multiple of 64");
   TORCH_CHECK(D == 64 || D == 128 || D == 256);
-  TORCH_CHECK(dout.is_contiguous() && q.is_contiguous() && k.is_contiguous() &&
-              v.is_contiguous() && o.is_contiguous());
+  TORCH_CHECK(dout.stride(3) == 1 && q.stride(3) == 1 && k.stride(3) == 1 &&
+              v.stride(3) == 1, "attn_bwd: innermost dim must be dense");
+  TORCH_CHECK(dout.stride(2) % 8 == 0 && q.stride(2) % 8 == 0 &&
+              k.stride(2) % 8 == 0 && v.stride(2) % 8 == 0);
 
   auto delta = (dout.to(at::kFloat) * o.to(at::kFloat)).sum(-1).contiguous();
-  auto dq_f32 = at::zeros({B, H, T, D}, q.options().dtype(at::kFloat));
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+  // grads physically [B, T, H, D] (matches the projection layout upstream,
+  // so the model-side transposes stay views)
+  auto dq_f32 = at::zeros({B, T, H, D}, q.options().dtype(at::kFloat))
+                    .permute({0, 2, 1, 3});
+  auto dk = at::empty({B, T, H, D}, k.options()).permute({0, 2, 1, 3});
+  auto dv = at::empty({B, T, H, D}, v.options()).permute({0, 2, 1, 3});
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
   dim3 grid(B * H, T / KB), block(256);
@@ -477,7 +515,9 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        dq_f32.data_ptr<float>(),                             \
                        reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
                        reinterpret_cast<unsigned short*>(dv.data_ptr()), T,  \
-                       scale, causal ? 1 : 0);                               \
+                       H, scale, causal ? 1 : 0, str_of(dout), str_of(q),    \
+                       str_of(k), str_of(v), str_of(dq_f32), str_of(dk),     \
+                       str_of(dv));                                          \
   } while (0)
 
   if (D == 64) LAUNCH_B(64);

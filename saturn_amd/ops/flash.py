@@ -34,7 +34,7 @@ class _FlashFn(torch.autograd.Function):
         ext = require_ext()
         if hasattr(ext, "attn_bwd"):
             dq, dk, dv = ext.attn_bwd(
-                do.contiguous(), q, k, v, o, lse, ctx.causal
+                _dense_rows(do), q, k, v, o, lse, ctx.causal
             )
             return dq, dk, dv, None
         # Analytic FA2 backward composed from rocBLAS GEMMs (recompute P
@@ -66,6 +66,15 @@ def _kernel_supported(q) -> bool:
     )
 
 
+def _dense_rows(t):
+    """The kernels address arbitrary (B, H, T) strides but need a dense,
+    16-B-aligned innermost dim — transposed views of the model's [B,T,H,D]
+    projections qualify, so no copies in the common path."""
+    if t.stride(-1) == 1 and t.stride(2) % 8 == 0:
+        return t
+    return t.contiguous()
+
+
 def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
     """q,k,v: [B, H, T, D] bf16/fp16 contiguous (GQA: H_kv may divide H)."""
     ext = require_ext()
@@ -75,7 +84,7 @@ def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
         return _FlashFn.apply(
-            q.contiguous(), k.contiguous(), v.contiguous(), causal
+            _dense_rows(q), _dense_rows(k), _dense_rows(v), causal
         )
     global _warned
     if not _warned:
