@@ -80,6 +80,14 @@ __global__ void norm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ 
 #pragma unroll
   for (int k = 0; k < CPT; ++k) { dwp[k] = 0.f; dbp[k] = 0.f; }
 
+  // Per-thread row cache (rows <= 4096 cols): the second pass (dx) reuses
+  // the values loaded by the reduction pass instead of re-reading dy/x
+  // from HBM (5 -> 3 global passes; the op is bandwidth-bound).  Longer
+  // rows would blow the register budget (CPT=64 -> 256 extra VGPRs), so
+  // they keep the re-read form.
+  constexpr bool CACHE = (CPT <= 16);
+  float c_dyw[CACHE ? CPT : 1];
+  float c_xh[CACHE ? CPT : 1];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + (long)row * cols;
     const T* xr = x + (long)row * cols;
@@ -91,6 +99,10 @@ __global__ void norm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ 
       float dyv = (float)dyr[j];
       float xh = ((float)xr[j] - mu) * rs;
       float dyw = dyv * (float)w[j];
+      if constexpr (CACHE) {
+        c_dyw[k] = dyw;
+        c_xh[k] = xh;
+      }
       s1 += dyw;
       s2 += dyw * xh;
       dwp[k] += dyv * xh;
@@ -98,10 +110,15 @@ __global__ void norm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ 
     }
     float m1 = RMS ? 0.f : block_sum<LN_BLOCK>(s1, lds) / cols;
     float m2 = block_sum<LN_BLOCK>(s2, lds) / cols;
-    for (int j = threadIdx.x; j < cols; j += LN_BLOCK) {
-      float dyw = (float)dyr[j] * (float)w[j];
-      float xh = ((float)xr[j] - mu) * rs;
-      dxr[j] = (T)(rs * (dyw - m1 - xh * m2));
+    if constexpr (CACHE) {
+      for (int k = 0, j = threadIdx.x; j < cols; j += LN_BLOCK, ++k)
+        dxr[j] = (T)(rs * (c_dyw[k] - m1 - c_xh[k] * m2));
+    } else {
+      for (int j = threadIdx.x; j < cols; j += LN_BLOCK) {
+        float dyw = (float)dyr[j] * (float)w[j];
+        float xh = ((float)xr[j] - mu) * rs;
+        dxr[j] = (T)(rs * (dyw - m1 - xh * m2));
+      }
     }
     __syncthreads();
   }
