@@ -314,7 +314,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
   const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
   const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
-  const unsigned short* dOh = dO + bh * (long)T * D;
+  const unsigned short* dOh = dO + b * dos.sb + h * dos.sh;
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
 
@@ -424,21 +424,22 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       }
     }
 
-    // ---- dQ partials: wave w handles q-subtile (w&1), d-half (w>>1)
+    // ---- dQ partials: wave w handles q-subtile (w&1), d-half (w>>1).
+    // NB: names must not shadow the head index h used in the atomic below.
     __syncthreads();
     {
-      const int n = wid & 1;
-      const int h = wid >> 1;
+      const int qsub = wid & 1;
+      const int dhalf = wid >> 1;
 #pragma unroll
       for (int dt2 = 0; dt2 < D / 32; ++dt2) {
-        const int dt = h * (D / 32) + dt2;
+        const int dt = dhalf * (D / 32) + dt2;
         f32x4 acc_dq = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kt2 = 0; kt2 < 2; ++kt2) {
           bf16x8 adst, bk;
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][n * 16 + r];
+            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
             bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
           }
           // A[q row][key k]: dst is [key][q]; the transposed read above
@@ -447,7 +448,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         }
 #pragma unroll
         for (int reg = 0; reg < 4; ++reg) {
-          const int q_glob = q0 + n * 16 + qg * 4 + reg;
+          const int q_glob = q0 + qsub * 16 + qg * 4 + reg;
           atomicAdd(&dQ[b * dqs.sb + h * dqs.sh + (long)q_glob * dqs.st +
                         dt * 16 + r],
                     acc_dq[reg]);
