@@ -17,7 +17,7 @@ datasets; BASELINE.md mandates synthetic + random init).
 
 from __future__ import annotations
 
-import math
+
 from dataclasses import dataclass
 
 import torch

@@ -8,7 +8,7 @@ numerics tests compare the two paths (tests/test_ops_gpu.py).
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

@@ -10,12 +10,12 @@ container.  The reference uses stock ``torch.optim.SGD``
 
 from __future__ import annotations
 
-import math
+
 from typing import Iterable, List, Optional
 
 import torch
 
-from saturn_amd.ops import has_ext, require_ext
+from saturn_amd.ops import require_ext
 
 
 def _grouped(params_with_grads: List[torch.nn.Parameter]):

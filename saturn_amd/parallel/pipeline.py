@@ -13,7 +13,7 @@ device overlap stage s of chunk j with stage s-1 of chunk j+1.
 
 from __future__ import annotations
 
-from typing import Callable, List, Optional, Sequence
+from typing import List, Optional, Sequence
 
 import torch
 import torch.nn as nn

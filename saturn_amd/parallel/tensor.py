@@ -13,7 +13,7 @@ receive identical gradients on every rank and need no extra sync.
 
 from __future__ import annotations
 
-from typing import List
+
 
 import torch
 import torch.distributed as dist
