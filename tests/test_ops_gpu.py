@@ -323,3 +323,40 @@ def test_flash_attention_strided_views_match_contiguous():
     out.backward(do.to(torch.bfloat16))
     e = rel_err(q.grad.transpose(1, 2), q32.grad)
     assert e < 5e-2, e
+
+
+def test_full_attention_noncausal_vs_fp32():
+    """BERT/ViT path: bidirectional flash kernel."""
+    from saturn_amd.ops import require_ext
+    from saturn_amd.ops.flash import flash_attention
+    from saturn_amd.ops.functional import attention_math
+
+    ext = require_ext()
+    if not hasattr(ext, "attn_fwd"):
+        pytest.skip("attn_fwd not built")
+    torch.manual_seed(0)
+    B, H, T, D = 2, 4, 256, 64
+    q32 = torch.randn(B, H, T, D, device="cuda")
+    k32 = torch.randn_like(q32)
+    v32 = torch.randn_like(q32)
+    ref = attention_math(q32, k32, v32, causal=False)
+    out = flash_attention(
+        q32.to(torch.bfloat16), k32.to(torch.bfloat16),
+        v32.to(torch.bfloat16), causal=False,
+    )
+    assert rel_err(out, ref) < 4e-2
+
+    # backward
+    q = q32.to(torch.bfloat16).requires_grad_(True)
+    k = k32.to(torch.bfloat16).requires_grad_(True)
+    v = v32.to(torch.bfloat16).requires_grad_(True)
+    out = flash_attention(q, k, v, causal=False)
+    do = torch.randn_like(out)
+    out.backward(do)
+    q32g = q32.detach().requires_grad_(True)
+    k32g = k32.detach().requires_grad_(True)
+    v32g = v32.detach().requires_grad_(True)
+    attention_math(q32g, k32g, v32g, causal=False).backward(do.float())
+    assert rel_err(q.grad, q32g.grad) < 5e-2
+    assert rel_err(k.grad, k32g.grad) < 5e-2
+    assert rel_err(v.grad, v32g.grad) < 5e-2

@@ -139,3 +139,29 @@ def test_fsdp_executor_world1_gpu(tmp_path):
     )
     params, bt = FSDPExecutor.search(t, [0], 940)
     assert params is not None and bt > 0
+
+
+def test_bert_and_vit_tiny_train_gpu():
+    from saturn_amd.models import get_bert_model, get_vit_model, mlm_loss, vit_loss
+    from saturn_amd.models.bert import SyntheticMLM
+    from saturn_amd.ops.optim import FusedAdam
+
+    torch.manual_seed(0)
+    bert = get_bert_model({"n_layer": 2, "n_embd": 256, "n_head": 4,
+                           "vocab_size": 1024, "n_ctx": 128}).to("cuda", torch.bfloat16)
+    ds = SyntheticMLM(4, 128, 1024)
+    x = torch.stack([ds[i][0] for i in range(4)]).cuda()
+    lab = torch.stack([ds[i][1] for i in range(4)]).cuda()
+    opt = FusedAdam(bert.parameters(), lr=1e-3)
+    loss = mlm_loss(bert(x), lab)
+    loss.backward()
+    opt.step()
+    assert torch.isfinite(torch.tensor(float(loss)))
+
+    vit = get_vit_model({"n_layer": 2, "n_embd": 256, "n_head": 4,
+                         "img_size": 64}).to("cuda", torch.bfloat16)
+    px = torch.randn(2, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    y = torch.randint(0, 1000, (2,), device="cuda")
+    loss = vit_loss(vit(px), y)
+    loss.backward()
+    assert torch.isfinite(torch.tensor(float(loss)))
