@@ -123,17 +123,24 @@ def execute(
     plan,
     task_dependency_dict: Optional[Dict] = None,
     launch_timeout: Optional[float] = None,
-) -> None:
+    raise_on_failure: bool = True,
+) -> List:
     """Execute one interval's tasks with MILP-ordered gang placement.
 
     Parity with reference ``executor.execute`` (executor.py:88-129): every
     relevant task is launched on its plan-assigned GPU set once all earlier
     tasks sharing any of its GPUs have finished; afterwards the parent
     advances each task's dataloader cursor.
+
+    Returns the list of tasks that FAILED this interval.  With
+    ``raise_on_failure=False`` the orchestrator can keep the rest of the
+    batch running and retry failures from their last checkpoint (elastic
+    recovery the reference lacks — a crash there kills the whole batch,
+    SURVEY §5.3).
     """
     n = len(relevant_tasks)
     if n == 0:
-        return
+        return []
     done_events = [threading.Event() for _ in range(n)]
     errors: List[Optional[str]] = [None] * n
 
@@ -200,12 +207,21 @@ def execute(
     for t in threads:
         t.join()
 
-    failed = [
-        (relevant_tasks[i].name, e) for i, e in enumerate(errors) if e is not None
-    ]
-    if failed:
-        details = "\n".join(f"--- {nm} ---\n{e}" for nm, e in failed)
-        raise RuntimeError(f"{len(failed)} task(s) failed this interval:\n{details}")
+    failed_idx = [i for i, e in enumerate(errors) if e is not None]
+    if failed_idx:
+        details = "\n".join(
+            f"--- {relevant_tasks[i].name} ---\n{errors[i]}" for i in failed_idx
+        )
+        if raise_on_failure:
+            raise RuntimeError(
+                f"{len(failed_idx)} task(s) failed this interval:\n{details}"
+            )
+        log.warning(
+            "%d task(s) failed this interval (will retry from checkpoint):\n%s",
+            len(failed_idx),
+            details,
+        )
 
     elapsed = timer() - start
     log.info("interval done: intended %.1fs actual %.1fs", interval, elapsed)
+    return [relevant_tasks[i] for i in failed_idx]

@@ -28,12 +28,18 @@ def orchestrate(
     solver_timeout: Optional[float] = None,
     hysteresis: float = 500.0,
     launch_timeout: Optional[float] = None,
+    max_task_retries: int = 2,
 ) -> None:
     """Run a batch of profiled tasks to completion.
 
     Parameters mirror the reference ``orchestrate(task_list, log, interval,
     gurobi)``; ``gurobi`` is gone (HiGHS is built in) and ``n_gpus`` pins the
     node size for CPU-only tests.
+
+    Elastic recovery: a task whose interval launch crashes is retried from
+    its last checkpoint in the next interval (up to ``max_task_retries``
+    times) while the rest of the batch keeps running — the reference aborts
+    the whole batch on any child failure.
     """
     if log_level:
         logging.basicConfig(
@@ -58,6 +64,7 @@ def orchestrate(
     apply_plan(task_list, plan)
 
     pool = ThreadPoolExecutor(max_workers=1)
+    fail_counts: dict = {}
     try:
         while task_list:
             relevant, batches, completing = forecast(task_list, interval, plan)
@@ -81,13 +88,46 @@ def orchestrate(
                 if next_tasks
                 else None
             )
-            execute(
+            failed = execute(
                 relevant,
                 batches,
                 interval,
                 plan,
                 launch_timeout=launch_timeout,
+                raise_on_failure=False,
             )
+            if failed:
+                for t in failed:
+                    fail_counts[t.name] = fail_counts.get(t.name, 0) + 1
+                    if fail_counts[t.name] > max_task_retries:
+                        raise RuntimeError(
+                            f"task {t.name} failed {fail_counts[t.name]} "
+                            "times; giving up on the batch"
+                        )
+                    # roll back this interval's forecast bookkeeping; the
+                    # checkpoint is the ground truth for progress
+                    idx = relevant.index(t)
+                    t.batches_completed = max(
+                        0, t.batches_completed - batches[idx]
+                    )
+                # failed-but-forecast-complete tasks must not retire; the
+                # overlapped solve used the stale retirement set, so drop it
+                # and re-solve synchronously on the corrected task list
+                if fut is not None:
+                    fut.cancel()
+                task_list = [
+                    t for t in task_list if t not in completing or t in failed
+                ]
+                plan = solve(
+                    task_list,
+                    None,
+                    interval=interval,
+                    timeout=solver_timeout,
+                    n_gpus=n_gpus,
+                    hysteresis=hysteresis,
+                )
+                apply_plan(task_list, plan)
+                continue
             task_list = next_tasks
             if fut is not None:
                 plan = fut.result()
