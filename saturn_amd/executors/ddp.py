@@ -100,6 +100,8 @@ def _ddp_worker(
             x, y = batch
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
+            if x.is_floating_point():
+                x = x.to(dtype)
             out = ddp(x)
             loss = task.loss_function(out, y)
             loss.backward()

@@ -59,6 +59,8 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             x, y = batch
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
+            if x.is_floating_point():
+                x = x.to(dtype)
             loss = task.loss_function(model(x), y)
             loss.backward()
             optimizer.step()

@@ -89,6 +89,8 @@ def _run_pipeline(task, n_devices: int, batch_count: int,
 
     def step(batch):
         x, y = batch
+        if x.is_floating_point():
+            x = x.to(dtype)
         out = pipe(x)
         loss = task.loss_function(out, y.to(last_dev, non_blocking=True))
         loss.backward()

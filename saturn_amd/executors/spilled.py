@@ -100,6 +100,8 @@ def _run_spilled(task, batch_count: int, params: Optional[Dict[str, Any]],
         x, y = batch
         x = x.to(device, non_blocking=True)
         y = y.to(device, non_blocking=True)
+        if x.is_floating_point():
+            x = x.to(dtype)
         loss = task.loss_function(z3(x), y)
         loss.backward()
         z3.grad_sync()

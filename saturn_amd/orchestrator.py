@@ -53,6 +53,18 @@ def orchestrate(
         solver_timeout = max(1.0, interval / 2)
 
     task_list = list(task_list)
+    infeasible = [
+        t for t in task_list
+        if not any(s is not None and s.feasible for s in t.strategies.values())
+    ]
+    if infeasible:
+        log.error(
+            "dropping task(s) with no feasible strategy (every trial cell "
+            "failed): %s", [t.name for t in infeasible],
+        )
+        task_list = [t for t in task_list if t not in infeasible]
+    if not task_list:
+        return
     plan: Plan = solve(
         task_list,
         None,
