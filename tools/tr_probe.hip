@@ -22,11 +22,15 @@ __global__ void k(short* out, int hyp) {
   unsigned off;
   if (hyp == 0) off = ((l & 15) + (l >> 4) * 64) * 2;      // canonical
   else off = (l & 15) * 2;                                  // no quadrant term
-  const unsigned short* addr = &img[0] + off / 2;
+  // ds ops take a 32-bit LDS byte address: addrspacecast to AS(3), truncate
+  unsigned a32 =
+      (unsigned)(unsigned long long)(__attribute__((address_space(3)))
+                                         unsigned short*)&img[0] +
+      off;
   short4v v;
   asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                : "=&v"(v)
-               : "v"(addr)
+               : "v"(a32)
                : "memory");
   for (int j = 0; j < 4; ++j) out[l * 4 + j] = v[j];
 }
