@@ -54,6 +54,45 @@ static TStr str_of(const at::Tensor& t) {
 // owned by quadrant r>>2 at accumulator reg r&3; its 16 score columns live
 // in that quadrant's 16 lanes, so row max/sum reduce with 4 shfl_xor steps
 // and never touch LDS.
+
+// ---------------------------------------------------------------------------
+// ds_read_b64_tr_b16 B-fragment read (hardware transpose).
+// Semantics pinned by tools/tr_probe.hip on MI355X: with per-lane source
+// address a(s), lane l's element j comes from element (l&3) of the aligned
+// 4-element window of lane s = (l&~15) + 4*j + ((l>>2)&3) in its 16-lane
+// group.  For the MFMA B-fragment B[(l>>4)*8+j][dcol=l&15] of a row-major
+// LDS tile, each lane addresses row (g*8 + ((l>>2)&3)), col
+// (col0 + 4*(l&3)) — two reads (rows +0, +4) yield the 8-slot fragment
+// with no bank-conflict-prone scalar gathers.  Addresses must be 8-B
+// aligned (row strides here are multiples of 8 B — G17).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) short bf16x4;
+
+__device__ __forceinline__ unsigned lds_addr32(const unsigned short* p) {
+  return (unsigned)(unsigned long long)(
+      __attribute__((address_space(3))) const unsigned short*)p;
+}
+
+template <int ROWPITCH>
+__device__ __forceinline__ bf16x8 tr16_bfrag(const unsigned short* tile_row0,
+                                             int lane) {
+  // tile_row0: &tile[base_row][col0] for this fragment's 8-row block.
+  const int src_row = (lane >> 2) & 3;
+  const int src_col = 4 * (lane & 3);
+  const unsigned a0 =
+      lds_addr32(tile_row0 + src_row * ROWPITCH + src_col);
+  const unsigned a1 = a0 + 4u * ROWPITCH * 2u;
+  bf16x4 v0, v1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(v0), "=&v"(v1)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  return __builtin_shufflevector(v0, v1, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
 template <int D>
 struct AttnLds {
   unsigned short k[KBLK][D + 8];
@@ -175,10 +214,7 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
     bf16x8 frag_p = *reinterpret_cast<const bf16x8*>(&lds.p[wid][r][qg * 8]);
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
-      bf16x8 bv;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        bv[j] = (short)lds.v[qg * 8 + j][dt * 16 + r];
+      bf16x8 bv = tr16_bfrag<D + 8>(&lds.v[qg * 8][dt * 16], lane);
       otile[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(frag_p, bv, otile[dt], 0, 0, 0);
     }
   }
@@ -411,12 +447,8 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
           *reinterpret_cast<const bf16x8*>(&lds.dst[wid * 16 + r][qg * 8]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
-        bf16x8 bdo, bq;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          bdo[j] = (short)lds.dot[qg * 8 + j][dt * 16 + r];
-          bq[j] = (short)lds.qt[qg * 8 + j][dt * 16 + r];
-        }
+        bf16x8 bdo = tr16_bfrag<D + 8>(&lds.dot[qg * 8][dt * 16], lane);
+        bf16x8 bq = tr16_bfrag<D + 8>(&lds.qt[qg * 8][dt * 16], lane);
         acc_dv[dt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bdo, acc_dv[dt], 0, 0, 0);
         acc_dk[dt] =
@@ -436,12 +468,10 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         f32x4 acc_dq = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kt2 = 0; kt2 < 2; ++kt2) {
-          bf16x8 adst, bk;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
-            bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
-          }
+          bf16x8 adst = tr16_bfrag<QT + 8>(
+              &lds.dst[kt2 * 32 + qg * 8][qsub * 16], lane);
+          bf16x8 bk =
+              tr16_bfrag<D + 8>(&lds.kt[kt2 * 32 + qg * 8][dt * 16], lane);
           // A[q row][key k]: dst is [key][q]; the transposed read above
           // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
           acc_dq = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bk, acc_dq, 0, 0, 0);

@@ -20,8 +20,10 @@ __global__ void k(short* out, int hyp) {
   __syncthreads();
   const int l = threadIdx.x;
   unsigned off;
-  if (hyp == 0) off = ((l & 15) + (l >> 4) * 64) * 2;      // canonical
-  else off = (l & 15) * 2;                                  // no quadrant term
+  if (hyp == 0) off = ((l & 15) + (l >> 4) * 64) * 2;      // linear in lane
+  else if (hyp == 1) off = (((l & 15) * 3) % 16 + (l >> 4) * 64) * 2;  // permuted
+  else if (hyp == 2) off = 128 * 2 * (l >> 4);             // uniform per group
+  else off = ((l & 15) * 16 + (l >> 4) * 512) * 2 % 2048;  // stride-16 rows
   // ds ops take a 32-bit LDS byte address: addrspacecast to AS(3), truncate
   unsigned a32 =
       (unsigned)(unsigned long long)(__attribute__((address_space(3)))
@@ -38,13 +40,13 @@ __global__ void k(short* out, int hyp) {
 int main() {
   short* d;
   hipMalloc(&d, 64 * 4 * 2);
-  for (int hyp = 0; hyp < 2; ++hyp) {
+  for (int hyp = 0; hyp < 4; ++hyp) {
     hipMemset(d, 0xff, 64 * 4 * 2);
     hipLaunchKernelGGL(k, dim3(1), dim3(64), 0, 0, d, hyp);
     short h[256];
     hipMemcpy(h, d, sizeof(h), hipMemcpyDeviceToHost);
     printf("hyp %d:\n", hyp);
-    for (int l = 0; l < 64; l += 1) {
+    for (int l = 0; l < 32; l += 1) {
       printf("  l%02d:", l);
       for (int j = 0; j < 4; ++j) printf(" %4d", h[l * 4 + j]);
       printf("\n");
