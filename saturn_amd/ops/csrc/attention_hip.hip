@@ -73,6 +73,30 @@ __device__ __forceinline__ unsigned lds_addr32(const unsigned short* p) {
       __attribute__((address_space(3))) const unsigned short*)p;
 }
 
+template <int P1, int P2>
+__device__ __forceinline__ void tr16_bfrag2(const unsigned short* t1_row0,
+                                            const unsigned short* t2_row0,
+                                            int lane, bf16x8& o1, bf16x8& o2) {
+  const int src_row = (lane >> 2) & 3;
+  const int src_col = 4 * (lane & 3);
+  const unsigned a0 = lds_addr32(t1_row0 + src_row * P1 + src_col);
+  const unsigned a1 = a0 + 4u * P1 * 2u;
+  const unsigned b0 = lds_addr32(t2_row0 + src_row * P2 + src_col);
+  const unsigned b1 = b0 + 4u * P2 * 2u;
+  bf16x4 v0, v1, w0, w1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(v0), "=&v"(v1), "=&v"(w0), "=&v"(w1)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+      : "memory");
+  o1 = __builtin_shufflevector(v0, v1, 0, 1, 2, 3, 4, 5, 6, 7);
+  o2 = __builtin_shufflevector(w0, w1, 0, 1, 2, 3, 4, 5, 6, 7);
+}
+
 template <int ROWPITCH>
 __device__ __forceinline__ bf16x8 tr16_bfrag(const unsigned short* tile_row0,
                                              int lane) {
@@ -447,8 +471,9 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
           *reinterpret_cast<const bf16x8*>(&lds.dst[wid * 16 + r][qg * 8]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
-        bf16x8 bdo = tr16_bfrag<D + 8>(&lds.dot[qg * 8][dt * 16], lane);
-        bf16x8 bq = tr16_bfrag<D + 8>(&lds.qt[qg * 8][dt * 16], lane);
+        bf16x8 bdo, bq;
+        tr16_bfrag2<D + 8, D + 8>(&lds.dot[qg * 8][dt * 16],
+                                  &lds.qt[qg * 8][dt * 16], lane, bdo, bq);
         acc_dv[dt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bdo, acc_dv[dt], 0, 0, 0);
         acc_dk[dt] =
@@ -468,10 +493,10 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         f32x4 acc_dq = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int kt2 = 0; kt2 < 2; ++kt2) {
-          bf16x8 adst = tr16_bfrag<QT + 8>(
-              &lds.dst[kt2 * 32 + qg * 8][qsub * 16], lane);
-          bf16x8 bk =
-              tr16_bfrag<D + 8>(&lds.kt[kt2 * 32 + qg * 8][dt * 16], lane);
+          bf16x8 adst, bk;
+          tr16_bfrag2<QT + 8, D + 8>(
+              &lds.dst[kt2 * 32 + qg * 8][qsub * 16],
+              &lds.kt[kt2 * 32 + qg * 8][dt * 16], lane, adst, bk);
           // A[q row][key k]: dst is [key][q]; the transposed read above
           // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
           acc_dq = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bk, acc_dq, 0, 0, 0);
