@@ -469,11 +469,17 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       bf16x8 ap = *reinterpret_cast<const bf16x8*>(&lds.pt[wid][r][qg * 8]);
       bf16x8 adst =
           *reinterpret_cast<const bf16x8*>(&lds.dst[wid * 16 + r][qg * 8]);
+      // NB: scalar gathers here beat tr16 reads — the asm's memory clobber
+      // would drain the freshly-issued pt/dst scalar writes at every step,
+      // while hipcc schedules these reads with counted waits (measured).
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         bf16x8 bdo, bq;
-        tr16_bfrag2<D + 8, D + 8>(&lds.dot[qg * 8][dt * 16],
-                                  &lds.qt[qg * 8][dt * 16], lane, bdo, bq);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          bdo[j] = (short)lds.dot[qg * 8 + j][dt * 16 + r];
+          bq[j] = (short)lds.qt[qg * 8 + j][dt * 16 + r];
+        }
         acc_dv[dt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bdo, acc_dv[dt], 0, 0, 0);
         acc_dk[dt] =
@@ -494,9 +500,11 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
 #pragma unroll
         for (int kt2 = 0; kt2 < 2; ++kt2) {
           bf16x8 adst, bk;
-          tr16_bfrag2<QT + 8, D + 8>(
-              &lds.dst[kt2 * 32 + qg * 8][qsub * 16],
-              &lds.kt[kt2 * 32 + qg * 8][dt * 16], lane, adst, bk);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
+            bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
+          }
           // A[q row][key k]: dst is [key][q]; the transposed read above
           // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
           acc_dq = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bk, acc_dq, 0, 0, 0);
