@@ -40,6 +40,11 @@ def main() -> None:
     ap.add_argument("--bucket-mb", type=float, default=64.0)
     args = ap.parse_args()
 
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        raise SystemExit(
+            "--gpus N>1 must be launched via torch.distributed.run "
+            "(one rank per GPU); a single process would hang at rendezvous."
+        )
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))

@@ -147,3 +147,31 @@ def test_remaining_work_shrinks_runtime(save_dir):
     t.batches_completed = 50
     plan = solve([t], n_gpus=1, timeout=10)
     assert plan.makespan == pytest.approx(50.0, rel=1e-3)
+
+
+def test_solver_scales_to_larger_batches(save_dir):
+    """12 tasks x 8 GPUs x 3 options each must solve (or greedy-fallback)
+    within the timeout and produce a valid plan (SURVEY §7 hard-part 4)."""
+    import random
+
+    rng = random.Random(7)
+    tasks = []
+    for i in range(12):
+        base = rng.uniform(20, 200)
+        tasks.append(
+            make_task(
+                f"s{i}",
+                {1: base, 2: base * 0.55, 4: base * 0.3},
+                save_dir,
+            )
+        )
+    import time
+
+    t0 = time.monotonic()
+    plan = solve(tasks, n_gpus=8, timeout=20)
+    elapsed = time.monotonic() - t0
+    assert elapsed < 60, f"solve took {elapsed:.1f}s"
+    check_plan_valid(plan, tasks, 8)
+    # must beat fully-sequential best-option schedule
+    seq = sum(min(s.runtime for s in t.strategies.values()) for t in tasks)
+    assert plan.makespan < seq
