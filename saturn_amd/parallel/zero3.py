@@ -61,6 +61,21 @@ def _reducescatter_into(shard: torch.Tensor, full: torch.Tensor):
     dist.reduce_scatter_tensor(shard, full)
 
 
+#: one copy stream per device, shared by all offload units — H2D prefetch
+#: of unit i+1 overlaps unit i's compute (double-buffered
+#: pinned-DRAM <-> HBM streaming on a dedicated HIP copy stream)
+_copy_streams: Dict[int, "torch.cuda.Stream"] = {}
+
+
+def _copy_stream(device) -> Optional["torch.cuda.Stream"]:
+    if device.type != "cuda":
+        return None
+    idx = device.index or 0
+    if idx not in _copy_streams:
+        _copy_streams[idx] = torch.cuda.Stream(device=device)
+    return _copy_streams[idx]
+
+
 class _Unit:
     """One shard group: a block's parameters flattened per dtype."""
 
@@ -106,23 +121,44 @@ class _Unit:
             p.data = self._placeholder
         self.full: Optional[torch.Tensor] = None
         self.work = None
+        self.copy_event: Optional["torch.cuda.Event"] = None
         self.grad_pending = 0
 
     # -- gather / free -----------------------------------------------------
     def start_gather(self, async_op: bool = False) -> None:
         if self.full is not None:
             return
-        shard_dev = (
-            self.shard.data.to(self.device, non_blocking=True)
-            if self.offload
-            else self.shard.data
-        )
+        cs = _copy_stream(self.device) if self.offload else None
+        if cs is not None:
+            # pinned-host -> HBM on the copy stream: a prefetched unit's
+            # upload overlaps the current unit's compute
+            self.full = torch.empty(
+                self.pad_numel, dtype=self.dtype, device=self.device
+            )
+            self.full.record_stream(cs)  # allocator: buffer is used on cs
+            ev = torch.cuda.Event()
+            with torch.cuda.stream(cs):
+                shard_dev = self.shard.data.to(self.device, non_blocking=True)
+                if _world() == 1:
+                    self.full.copy_(shard_dev)
+                    self.work = None
+                else:
+                    self.work = _allgather_into(
+                        self.full, shard_dev, async_op=async_op
+                    )
+                ev.record(cs)
+            self.copy_event = ev
+            return
+        shard_dev = self.shard.data
         self.full = torch.empty(
             self.pad_numel, dtype=self.dtype, device=self.device
         )
         self.work = _allgather_into(self.full, shard_dev, async_op=async_op)
 
     def finish_gather(self) -> None:
+        if self.copy_event is not None:
+            torch.cuda.current_stream(self.device).wait_event(self.copy_event)
+            self.copy_event = None
         if self.work is not None:
             self.work.wait()
             self.work = None
@@ -132,6 +168,10 @@ class _Unit:
     def free(self) -> None:
         for p in self.params:
             p.data = self._placeholder
+        if self.copy_event is not None:
+            # never drop a buffer with an in-flight upload
+            self.copy_event.synchronize()
+            self.copy_event = None
         self.full = None
         self.work = None
 
