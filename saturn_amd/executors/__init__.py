@@ -3,5 +3,6 @@ from .fsdp import FSDPExecutor
 from .pipeline import PipelineExecutor
 from .megatron import MegatronExecutor
 from .spilled import SpilledExecutor
+from .ulysses import UlyssesExecutor
 
-__all__ = ["DDPExecutor", "FSDPExecutor", "PipelineExecutor", "SpilledExecutor", "MegatronExecutor"]
+__all__ = ["DDPExecutor", "FSDPExecutor", "PipelineExecutor", "SpilledExecutor", "MegatronExecutor", "UlyssesExecutor"]

@@ -66,7 +66,7 @@ class GPTJAttention(nn.Module):
         q = self.q_proj(x).view(B, T, H, D)
         k = self.k_proj(x).view(B, T, H, D)
         v = self.v_proj(x).view(B, T, H, D)
-        cos, sin = self.rope_cos[:T], self.rope_sin[:T]
+        cos, sin = self.rope_cos, self.rope_sin  # apply_rope slices positions
         # partial rotary: the kernel rotates the first rotary_dim dims in
         # place and passes the rest through — no split+cat (the reference
         # materializes both halves and concatenates, GPTJ.py:255-259)

@@ -65,7 +65,7 @@ class LlamaAttention(nn.Module):
         q = self.q_proj(x).view(B, T, self.n_head, D)
         k = self.k_proj(x).view(B, T, self.n_kv, D)
         v = self.v_proj(x).view(B, T, self.n_kv, D)
-        cos, sin = self.rope_cos[:T], self.rope_sin[:T]
+        cos, sin = self.rope_cos, self.rope_sin  # apply_rope slices positions
         q = apply_rope(q, cos, sin, half_style=True)
         k = apply_rope(k, cos, sin, half_style=True)
         o = causal_attention(

@@ -211,6 +211,16 @@ def apply_rope(
     """
     B, T, H, D = x.shape
     half = cos.shape[-1]
+    from saturn_amd.parallel.sequence import _STATE as _sp
+
+    off = _sp["rank"] * T if _sp["world"] > 1 else 0
+    if cos.shape[0] < off + T:
+        raise ValueError(
+            "rope table too short for this sequence shard — pass the FULL "
+            "table to apply_rope (it slices positions itself)"
+        )
+    cos = cos[off : off + T]
+    sin = sin[off : off + T]
     # model.to(dtype=bf16) converts registered buffers — force fp32 tables
     cos_bt = cos.float().unsqueeze(0).expand(B, T, half).contiguous()
     sin_bt = sin.float().unsqueeze(0).expand(B, T, half).contiguous()
@@ -236,7 +246,7 @@ def attention_math(q, k, v, causal: bool = True) -> torch.Tensor:
     return torch.matmul(p, v.float()).to(q.dtype)
 
 
-def _attention(q, k, v, causal: bool) -> torch.Tensor:
+def _attention_core(q, k, v, causal: bool) -> torch.Tensor:
     if q.is_cuda:
         from saturn_amd.ops import flash  # local import: optional kernel
 
@@ -246,6 +256,16 @@ def _attention(q, k, v, causal: bool) -> torch.Tensor:
         k = k.repeat_interleave(rep, dim=1)
         v = v.repeat_interleave(rep, dim=1)
     return attention_math(q, k, v, causal=causal)
+
+
+def _attention(q, k, v, causal: bool) -> torch.Tensor:
+    from saturn_amd.parallel.sequence import sp_attention, sp_world
+
+    if sp_world() > 1:
+        # Ulysses: all-to-all seq-shard <-> head-shard around the full-
+        # sequence kernel (exact causal masking on the gathered sequence)
+        return sp_attention(q, k, v, lambda a, b, c: _attention_core(a, b, c, causal))
+    return _attention_core(q, k, v, causal)
 
 
 def causal_attention(q, k, v) -> torch.Tensor:
