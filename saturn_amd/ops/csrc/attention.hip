@@ -169,24 +169,40 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   float l_reg[4] = {0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? (q0_block + QBLK) : T;
+  // ---- async-stage split (guide T14): each thread holds the NEXT tile's
+  // K/V rows in registers; the global loads for tile i+1 are issued before
+  // tile i's compute so HBM latency hides under the MFMAs, and the LDS
+  // write happens right after the tile-consumed barrier.
+  constexpr int CHUNKS = (KBLK * D) / (256 * 8);
+  bf16x8 stg_k[CHUNKS], stg_v[CHUNKS];
+#pragma unroll
+  for (int c = 0; c < CHUNKS; ++c) {
+    const int idx = (c * 256 + threadIdx.x) * 8;
+    const int row = idx / D, col = idx % D;
+    stg_k[c] = *reinterpret_cast<const bf16x8*>(Kh + (long)row * ks.st + col);
+    stg_v[c] = *reinterpret_cast<const bf16x8*>(Vh + (long)row * vs.st + col);
+  }
   for (int kv0 = 0; kv0 < kv_end; kv0 += KBLK) {
-    // ---- stage K/V tile cooperatively (256 threads, coalesced rows)
-    __syncthreads();
-    {
-      // each thread copies (KBLK*D)/256 bf16 elements as 8-wide chunks
-      constexpr int CHUNKS = (KBLK * D) / (256 * 8);
+    __syncthreads();  // previous tile fully consumed
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int idx = (c * 256 + threadIdx.x) * 8;
+      const int row = idx / D, col = idx % D;
+      *reinterpret_cast<bf16x8*>(&lds.k[row][col]) = stg_k[c];
+      *reinterpret_cast<bf16x8*>(&lds.v[row][col]) = stg_v[c];
+    }
+    if (kv0 + KBLK < kv_end) {
+#pragma unroll
       for (int c = 0; c < CHUNKS; ++c) {
-        int idx = (c * 256 + threadIdx.x) * 8;
-        int row = idx / D, col = idx % D;
-        *reinterpret_cast<bf16x8*>(&lds.k[row][col]) =
-            *reinterpret_cast<const bf16x8*>(
-                Kh + (long)(kv0 + row) * ks.st + col);
-        *reinterpret_cast<bf16x8*>(&lds.v[row][col]) =
-            *reinterpret_cast<const bf16x8*>(
-                Vh + (long)(kv0 + row) * vs.st + col);
+        const int idx = (c * 256 + threadIdx.x) * 8;
+        const int row = idx / D, col = idx % D;
+        stg_k[c] = *reinterpret_cast<const bf16x8*>(
+            Kh + (long)(kv0 + KBLK + row) * ks.st + col);
+        stg_v[c] = *reinterpret_cast<const bf16x8*>(
+            Vh + (long)(kv0 + KBLK + row) * vs.st + col);
       }
     }
-    __syncthreads();
+    __syncthreads();  // staged tile visible
 
     // ---- S[16][32] = scale * Q K^T  (two 16-col MFMA tiles)
     f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
