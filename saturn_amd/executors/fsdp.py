@@ -66,6 +66,24 @@ def _fsdp_worker(
         holder = _ShardHolder(z3.sharded_parameters())
         optimizer = _make_optimizer(task, holder)
 
+        # Sharded optimizer-state resume (the reference loses optimizer
+        # moments at every interval; full-state ckpts carry
+        # them for DDP — here each rank reloads its own shard state when
+        # the solver kept the same world size, the common case under plan
+        # hysteresis).
+        import os as _os
+
+        opt_path = _os.path.join(
+            task.save_dir, f"{task.name}.optshard.w{world}.r{rank}.pt"
+        )
+        if not trial and _os.path.isfile(opt_path):
+            try:
+                optimizer.load_state_dict(
+                    torch.load(opt_path, map_location="cpu", weights_only=False)
+                )
+            except Exception:
+                pass  # layout changed; fresh moments
+
         it = task.get_iterator() if not trial else task.get_fresh_iterator()
 
         def next_batch():
@@ -107,6 +125,10 @@ def _fsdp_worker(
             sd = z3.full_state_dict()
             if rank == 0 and sd is not None:
                 task.save_checkpoint(sd, None)
+            # every rank persists its optimizer shard state
+            tmp = opt_path + ".tmp"
+            torch.save(optimizer.state_dict(), tmp)
+            _os.replace(tmp, opt_path)
             import torch.distributed as dist
 
             if world > 1:

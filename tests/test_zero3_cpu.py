@@ -146,3 +146,34 @@ def test_fsdp_executor_search_and_execute(save_dir, library_path):
     t.select_strategy(t.strategies[2])
     FSDPExecutor.execute(t, [0, 1], 911, 2)
     assert t.has_ckpt()
+
+
+def test_fsdp_optimizer_shard_resume(save_dir, library_path):
+    """Second interval resumes optimizer shard state (same world size)."""
+    import os
+
+    t = Task(
+        get_mlp_model,
+        get_mlp_dataloader,
+        mse_loss,
+        HParams(lr=1e-2, batch_count=6,
+                optimizer_cls=__import__("torch").optim.Adam),
+        gpu_range=[1],
+        name="fsdp_opt",
+        save_dir=save_dir,
+    )
+    params, bt = FSDPExecutor.search(t, [0], 915)
+    assert params is not None
+    t.strategies[1] = __import__("saturn_amd").Strategy(
+        FSDPExecutor, 1, params, bt * 6, batch_time=bt
+    )
+    t.select_strategy(t.strategies[1])
+    FSDPExecutor.execute(t, [0], 915, 3)
+    shard_file = os.path.join(save_dir, "fsdp_opt.optshard.w1.r0.pt")
+    assert os.path.isfile(shard_file)
+    import torch
+
+    st = torch.load(shard_file, weights_only=False)
+    assert any("exp_avg" in v for v in st["state"].values())
+    FSDPExecutor.execute(t, [0], 915, 3)  # second interval loads it back
+    assert t.has_ckpt()
