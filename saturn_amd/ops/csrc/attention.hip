@@ -277,6 +277,246 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   }
 }
 
+// ===========================================================================
+// Forward v2 (opt-in: SAMD_ATTN_V2=1): the CDNA4-guide structure — swapped
+// QK^T on v_mfma_f32_32x32x16_bf16 so each lane's 16 score values belong to
+// ONE q row (lane-local online softmax, one shfl_xor(32) merge), P stays in
+// registers and reaches the PV A-fragment via permlane32_swap half
+// exchanges; V rides the tr16 transpose read.  2x the q rows per MFMA of
+// the v1 16x16 structure.  Layouts validated by tools/mfma_probe.hip:
+//   32x32x16  A[m][k]: lane l, j -> A[l&31][(l>>5)*8+j]
+//             B[k][n]: lane l, j -> B[(l>>5)*8+j][l&31]
+//             D[m][n]: lane l, reg -> D[(reg&3)+8*(reg>>2)+4*(l>>5)][l&31]
+// ===========================================================================
+
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+template <int D>
+struct AttnV2Lds {
+  unsigned short k[KBLK][D + 8];
+  unsigned short v[KBLK][D + 8];
+};
+
+template <int D>
+__launch_bounds__(256, 2)
+__global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
+                                   const unsigned short* __restrict__ K,
+                                   const unsigned short* __restrict__ V,
+                                   unsigned short* __restrict__ O,
+                                   float* __restrict__ LSE, int T,
+                                   int n_heads, float scale, int causal,
+                                   TStr qs, TStr ks, TStr vs, TStr os) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  AttnV2Lds<D>& lds = *reinterpret_cast<AttnV2Lds<D>*>(smem);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int qr = lane & 31;       // this lane's q row (within the wave's 32)
+  const int hi = lane >> 5;       // half index
+  const long bh = blockIdx.x;
+  const long b = bh / n_heads, h = bh % n_heads;
+  const int q0_block = blockIdx.y * (4 * 32);
+  const int q0 = q0_block + wid * 32;
+  const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
+
+  // Q fragments (B operand of swapped QK^T): frag_q[ds][j] =
+  // Q[q0+qr][ds*16 + hi*8 + j].  At D=256 the preloaded array would push
+  // the kernel to 256 VGPRs with spills; the QK loop re-reads Q from L2
+  // per tile instead (the wave's 16 KB stays hot).
+  constexpr bool QREG = (D < 256);
+  bf16x8 frag_q[QREG ? D / 16 : 1];
+  if constexpr (QREG) {
+#pragma unroll
+    for (int ds = 0; ds < D / 16; ++ds)
+      frag_q[ds] = *reinterpret_cast<const bf16x8*>(
+          Qh + (long)(q0 + qr) * qs.st + ds * 16 + hi * 8);
+  }
+
+  // O accumulators: oacc[dv][reg] = O[(reg&3)+8*(reg>>2)+4*hi][dv*32 + qr]
+  f32x16 oacc[D / 32];
+#pragma unroll
+  for (int dv = 0; dv < D / 32; ++dv)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) oacc[dv][e] = 0.f;
+
+  float m_row = -INFINITY, l_row = 0.f;
+
+  // staging registers (T14 split)
+  constexpr int CHUNKS = (KBLK * D) / (256 * 8);
+  bf16x8 stg_k[CHUNKS], stg_v[CHUNKS];
+#pragma unroll
+  for (int c = 0; c < CHUNKS; ++c) {
+    const int idx = (c * 256 + threadIdx.x) * 8;
+    const int row = idx / D, col = idx % D;
+    stg_k[c] = *reinterpret_cast<const bf16x8*>(Kh + (long)row * ks.st + col);
+    stg_v[c] = *reinterpret_cast<const bf16x8*>(Vh + (long)row * vs.st + col);
+  }
+
+  const int kv_end = causal ? (q0_block + 4 * 32) : T;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KBLK) {
+    __syncthreads();
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      const int idx = (c * 256 + threadIdx.x) * 8;
+      const int row = idx / D, col = idx % D;
+      *reinterpret_cast<bf16x8*>(&lds.k[row][col]) = stg_k[c];
+      *reinterpret_cast<bf16x8*>(&lds.v[row][col]) = stg_v[c];
+    }
+    if (kv0 + KBLK < kv_end) {
+#pragma unroll
+      for (int c = 0; c < CHUNKS; ++c) {
+        const int idx = (c * 256 + threadIdx.x) * 8;
+        const int row = idx / D, col = idx % D;
+        stg_k[c] = *reinterpret_cast<const bf16x8*>(
+            Kh + (long)(kv0 + KBLK + row) * ks.st + col);
+        stg_v[c] = *reinterpret_cast<const bf16x8*>(
+            Vh + (long)(kv0 + KBLK + row) * vs.st + col);
+      }
+    }
+    __syncthreads();
+
+    // ---- S^T[key][q] = K Q^T: one 32x32 tile over the D contraction
+    f32x16 st;
+#pragma unroll
+    for (int e = 0; e < 16; ++e) st[e] = 0.f;
+#pragma unroll
+    for (int ds = 0; ds < D / 16; ++ds) {
+      bf16x8 ak = *reinterpret_cast<const bf16x8*>(
+          &lds.k[qr][ds * 16 + hi * 8]);
+      bf16x8 fq;
+      if constexpr (QREG)
+        fq = frag_q[ds];
+      else
+        fq = *reinterpret_cast<const bf16x8*>(
+            Qh + (long)(q0 + qr) * qs.st + ds * 16 + hi * 8);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ak, fq, st, 0, 0, 0);
+    }
+
+    // ---- lane-local online softmax for q row (q0 + qr).
+    // lane holds keys key16(reg) = (reg&3) + 8*(reg>>2) + 4*hi.
+    const int q_glob = q0 + qr;
+    float p[16];
+    float tile_max = -INFINITY;
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int key_loc = (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+      float sv = st[reg] * scale;
+      if (causal && kv0 + key_loc > q_glob) sv = -INFINITY;
+      p[reg] = sv;
+      tile_max = fmaxf(tile_max, sv);
+    }
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+    const float m_new = fmaxf(m_row, tile_max);
+    const float alpha = (m_row == -INFINITY) ? 0.f : __expf(m_row - m_new);
+    float psum = 0.f;
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const float pv = (p[reg] == -INFINITY) ? 0.f : __expf(p[reg] - m_new);
+      p[reg] = pv;
+      psum += pv;
+    }
+    psum += __shfl_xor(psum, 32, 64);
+    l_row = l_row * alpha + psum;
+    m_row = m_new;
+
+    // ---- P (f32, keys in D-layout order) -> PV A-fragments via pack +
+    // permlane32_swap.  pk[i] packs keys (2i, 2i+1) of this lane's set:
+    //   lo half (hi=0): keys {0,1},{2,3},{8,9},{10,11},{16,17},...
+    //   hi half (hi=1): keys {4,5},{6,7},{12,13},{14,15},...
+    // after swap(pk[2g], pk[2g+1]? see below) each lane holds the pairs of
+    // ITS A-fragment rows: frag keys (l>>5)*8 + 0..7 per 16-key tile.
+    unsigned pk[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const unsigned lo = f2us(p[2 * i]);
+      const unsigned hi2 = f2us(p[2 * i + 1]);
+      pk[i] = lo | (hi2 << 16);
+    }
+    // tile g (g=0: keys 0-15, g=1: keys 16-31): lane pairs pk[4g+0..3] hold
+    //   hi=0: (0,1),(2,3),(8,9),(10,11)   [+16 for g=1]
+    //   hi=1: (4,5),(6,7),(12,13),(14,15)
+    // swap(pk[4g+0], pk[4g+2]): lo lane gets {(0,1) | (4,5)->? }
+    bf16x8 pa[2];
+#pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      auto r0 = __builtin_amdgcn_permlane32_swap(pk[4 * g + 0], pk[4 * g + 2],
+                                                 false, false);
+      auto r1 = __builtin_amdgcn_permlane32_swap(pk[4 * g + 1], pk[4 * g + 3],
+                                                 false, false);
+      // r0[0]: lo half keeps (0,1), hi half got lo's (8,9)
+      // r0[1]: lo half got hi's (4,5), hi half keeps (12,13)
+      // r1[0]: (2,3) / lo's (10,11);  r1[1]: hi's (6,7) / (14,15)
+      // A-frag for this lane: keys hi*8 + {0..7} =
+      //   lo: (0,1),(2,3),(4,5),(6,7) = r0[0], r1[0], r0[1], r1[1]
+      //   hi: (8,9),(10,11),(12,13),(14,15) = r0[0], r1[0], r0[1], r1[1]
+      unsigned w0 = r0[0], w1 = r1[0], w2 = r0[1], w3 = r1[1];
+      pa[g][0] = (short)(w0 & 0xffff);
+      pa[g][1] = (short)(w0 >> 16);
+      pa[g][2] = (short)(w1 & 0xffff);
+      pa[g][3] = (short)(w1 >> 16);
+      pa[g][4] = (short)(w2 & 0xffff);
+      pa[g][5] = (short)(w2 >> 16);
+      pa[g][6] = (short)(w3 & 0xffff);
+      pa[g][7] = (short)(w3 >> 16);
+    }
+
+    // ---- rescale O by alpha of each accumulated row (rows differ from
+    // this lane's softmax row): alpha[row] sits in lanes row and row+32.
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int row = (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+      const float al = __shfl(alpha, row, 64);
+#pragma unroll
+      for (int dv = 0; dv < D / 32; ++dv) oacc[dv][reg] *= al;
+    }
+
+    // ---- PV: O^T? no — O[q][d] += P[q][k] V[k][d]:
+    // mfma(A=pa (32 q x 16 k), B=V[k][d] 16x32) per (key half g, dv)
+#pragma unroll
+    for (int dv = 0; dv < D / 32; ++dv) {
+#pragma unroll
+      for (int g = 0; g < 2; ++g) {
+        // B fragment via tr16: B[kk=hi*8+j][dcol=l&31]; group-dependent
+        // base col handled through the per-lane source address
+        const int src_row = g * 16 + ((lane >> 4) & 2) * 4 + ((lane >> 2) & 3);
+        const int src_col = dv * 32 + ((lane >> 4) & 1) * 16 + 4 * (lane & 3);
+        const unsigned a0 = lds_addr32(&lds.v[src_row][src_col]);
+        const unsigned a1 = a0 + 4u * (D + 8) * 2u;
+        bf16x4 v0, v1;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %2\n\t"
+            "ds_read_b64_tr_b16 %1, %3\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(v0), "=&v"(v1)
+            : "v"(a0), "v"(a1)
+            : "memory");
+        bf16x8 bv = __builtin_shufflevector(v0, v1, 0, 1, 2, 3, 4, 5, 6, 7);
+        oacc[dv] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[g], bv,
+                                                           oacc[dv], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue
+  unsigned short* Oh = O + b * os.sb + h * os.sh;
+  const float linv_own = 1.f / l_row;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+    const float linv = __shfl(linv_own, row, 64);
+#pragma unroll
+    for (int dv = 0; dv < D / 32; ++dv)
+      Oh[(long)(q0 + row) * os.st + dv * 32 + qr] = f2us(oacc[dv][reg] * linv);
+  }
+  if (hi == 0 && LSE != nullptr)
+    LSE[bh * (long)T + q0 + qr] = m_row + __logf(l_row);
+}
+
+
+
 // ---------------------------------------------------------------------------
 // Host wrapper
 // ---------------------------------------------------------------------------
@@ -303,6 +543,32 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
+
+  // opt-in v2 structure (swapped 32x32 MFMAs, in-register P)
+  static const bool use_v2 = [] {
+    const char* e = getenv("SAMD_ATTN_V2");
+    return e && e[0] == '1';
+  }();
+  // D=256 would spill at this structure (256 VGPRs + 53); v1 handles it
+  if (use_v2 && T % 128 == 0 && D <= 128) {
+    dim3 grid2(B * H, T / 128), block2(256);
+#define LAUNCH_V2(DD)                                                        \
+    do {                                                                     \
+      size_t shmem = sizeof(AttnV2Lds<DD>);                                  \
+      hipLaunchKernelGGL((attn_fwd_v2_kernel<DD>), grid2, block2, shmem,     \
+                         stream.stream(),                                    \
+                         reinterpret_cast<const unsigned short*>(q.data_ptr()), \
+                         reinterpret_cast<const unsigned short*>(k.data_ptr()), \
+                         reinterpret_cast<const unsigned short*>(v.data_ptr()), \
+                         reinterpret_cast<unsigned short*>(o.data_ptr()),    \
+                         lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0, \
+                         str_of(q), str_of(k), str_of(v), str_of(o));        \
+    } while (0)
+    if (D == 64) LAUNCH_V2(64);
+    else LAUNCH_V2(128);
+#undef LAUNCH_V2
+    return {o, lse};
+  }
   dim3 grid(B * H, T / QBLK), block(256);
 
 #define LAUNCH(DD)                                                          \
@@ -658,3 +924,4 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
 }
 
 }  // namespace samd
+
