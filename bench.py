@@ -35,7 +35,8 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--batch", type=int, default=16, help="per-GPU batch size")
     ap.add_argument("--seq", type=int, default=512)
-    ap.add_argument("--layers", type=int, default=28)
+    ap.add_argument("--layers", type=int, default=None,
+                    help="override layer count (default: model standard)")
     ap.add_argument("--model", type=str, default="gptj-6b")
     ap.add_argument("--bucket-mb", type=float, default=64.0)
     args = ap.parse_args()
@@ -68,17 +69,41 @@ def main() -> None:
 
         require_ext()
 
-    from saturn_amd.models.gptj import GPTJConfig, GPTJForCausalLM, pretraining_loss
     from saturn_amd.ops.optim import FusedSGD
     from saturn_amd.parallel.ddp import BucketedDDP
 
-    cfg = GPTJConfig(n_layer=args.layers, n_ctx=args.seq)
     torch.manual_seed(1234)
-    log(f"[rank {rank}] building GPT-J ({cfg.n_layer} layers) on {device}...")
+    log(f"[rank {rank}] building {args.model} on {device}...")
     # build directly on the device: at world 8 a host-side build would
     # transiently hold 8 x 24 GB fp32 replicas in DRAM
     with device:
-        model = GPTJForCausalLM(cfg)
+        if args.model == "llama-3-8b":
+            from saturn_amd.models.llama import (
+                LlamaConfig,
+                LlamaForCausalLM,
+                llama_loss as loss_fn,
+            )
+
+            mcfg = LlamaConfig(n_ctx=max(args.seq, 2048))
+            if args.layers is not None:
+                from dataclasses import replace
+
+                mcfg = replace(mcfg, n_layer=args.layers)
+            model = LlamaForCausalLM(mcfg)
+            vocab = mcfg.vocab_size
+        else:
+            from saturn_amd.models.gptj import (
+                GPTJConfig,
+                GPTJForCausalLM,
+                pretraining_loss as loss_fn,
+            )
+
+            mcfg = GPTJConfig(
+                n_layer=args.layers if args.layers is not None else 28,
+                n_ctx=args.seq,
+            )
+            model = GPTJForCausalLM(mcfg)
+            vocab = mcfg.vocab_size
     model = model.to(dtype=dtype)
     model.train()
     n_params = sum(p.numel() for p in model.parameters())
@@ -86,11 +111,11 @@ def main() -> None:
     opt = FusedSGD(model.parameters(), lr=1e-5)
 
     g = torch.Generator(device="cpu").manual_seed(4321 + rank)
-    x = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), generator=g).to(device)
+    x = torch.randint(0, vocab, (args.batch, args.seq), generator=g).to(device)
 
     def step() -> None:
         logits = ddp(x)
-        loss = pretraining_loss(logits, x)
+        loss = loss_fn(logits, x)
         loss.backward()
         ddp.grad_sync()
         opt.step()
