@@ -130,8 +130,8 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
                                 const unsigned short* __restrict__ V,
                                 unsigned short* __restrict__ O,
                                 float* __restrict__ LSE, int T, int n_heads,
-                                float scale, int causal, TStr qs, TStr ks,
-                                TStr vs, TStr os) {
+                                int n_kv, float scale, int causal, TStr qs,
+                                TStr ks, TStr vs, TStr os) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnLds<D>& lds = *reinterpret_cast<AttnLds<D>*>(smem);
 
@@ -144,9 +144,10 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
   const int q0 = q0_block + wid * WROWS;         // wave's first q row
   const long bh = blockIdx.x;
   const long b = bh / n_heads, h = bh % n_heads;
+  const long h_kv = h * n_kv / n_heads;  // GQA: q heads share kv heads
   const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
-  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
-  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h_kv * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h_kv * vs.sh;
 
   const int r = lane & 15;       // fragment row/col index
   const int qg = lane >> 4;      // quadrant 0..3
@@ -305,8 +306,9 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
                                    const unsigned short* __restrict__ V,
                                    unsigned short* __restrict__ O,
                                    float* __restrict__ LSE, int T,
-                                   int n_heads, float scale, int causal,
-                                   TStr qs, TStr ks, TStr vs, TStr os) {
+                                   int n_heads, int n_kv, float scale,
+                                   int causal, TStr qs, TStr ks, TStr vs,
+                                   TStr os) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnV2Lds<D>& lds = *reinterpret_cast<AttnV2Lds<D>*>(smem);
 
@@ -316,11 +318,12 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
   const int hi = lane >> 5;       // half index
   const long bh = blockIdx.x;
   const long b = bh / n_heads, h = bh % n_heads;
+  const long h_kv = h * n_kv / n_heads;  // GQA
   const int q0_block = blockIdx.y * (4 * 32);
   const int q0 = q0_block + wid * 32;
   const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
-  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
-  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h_kv * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h_kv * vs.sh;
 
   // Q fragments (B operand of swapped QK^T): frag_q[ds][j] =
   // Q[q0+qr][ds*16 + hi*8 + j].  At D=256 the preloaded array would push
@@ -529,6 +532,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
               "attn_fwd: innermost dim must be dense");
   const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
             D = (int)q.size(3);
+  const int Hkv = (int)k.size(1);
+  TORCH_CHECK(H % Hkv == 0, "attn_fwd: q heads must be a multiple of kv heads");
   TORCH_CHECK(k.size(2) == T, "attn_fwd: q/k length mismatch");
   TORCH_CHECK(T % QBLK == 0, "attn_fwd: T must be a multiple of 64");
   TORCH_CHECK(D == 64 || D == 128 || D == 256, "attn_fwd: D in {64,128,256}");
@@ -563,8 +568,9 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                          reinterpret_cast<const unsigned short*>(k.data_ptr()), \
                          reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                          reinterpret_cast<unsigned short*>(o.data_ptr()),    \
-                         lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0, \
-                         str_of(q), str_of(k), str_of(v), str_of(o));        \
+                         lse.data_ptr<float>(), T, H, Hkv, scale,            \
+                         causal ? 1 : 0, str_of(q), str_of(k), str_of(v),    \
+                         str_of(o));                                         \
     } while (0)
     if (D == 64) LAUNCH_V2(64);
     else LAUNCH_V2(128);
@@ -582,8 +588,9 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                        reinterpret_cast<const unsigned short*>(k.data_ptr()), \
                        reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                        reinterpret_cast<unsigned short*>(o.data_ptr()),     \
-                       lse.data_ptr<float>(), T, H, scale, causal ? 1 : 0,  \
-                       str_of(q), str_of(k), str_of(v), str_of(o));         \
+                       lse.data_ptr<float>(), T, H, Hkv, scale,             \
+                       causal ? 1 : 0, str_of(q), str_of(k), str_of(v),     \
+                       str_of(o));                                          \
   } while (0)
 
   if (D == 64) LAUNCH(64);
@@ -681,10 +688,12 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 const float* __restrict__ DELTA,
                                 float* __restrict__ dQ,
                                 unsigned short* __restrict__ dK,
-                                unsigned short* __restrict__ dV, int T,
-                                int n_heads, float scale, int causal,
-                                TStr dos, TStr qs, TStr ks, TStr vs,
-                                TStr dqs, TStr dks, TStr dvs) {
+                                unsigned short* __restrict__ dV,
+                                float* __restrict__ dKf,
+                                float* __restrict__ dVf, int T, int n_heads,
+                                int n_kv, float scale, int causal, TStr dos,
+                                TStr qs, TStr ks, TStr vs, TStr dqs,
+                                TStr dks, TStr dvs) {
   using namespace bwd;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnBwdLds<D>& lds = *reinterpret_cast<AttnBwdLds<D>*>(smem);
@@ -697,9 +706,10 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   const int kv0 = blockIdx.y * KB;
   const long bh = blockIdx.x;
   const long b = bh / n_heads, h = bh % n_heads;
+  const long h_kv = h * n_kv / n_heads;  // GQA: q heads share kv heads
   const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
-  const unsigned short* Kh = K + b * ks.sb + h * ks.sh;
-  const unsigned short* Vh = V + b * vs.sb + h * vs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h_kv * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h_kv * vs.sh;
   const unsigned short* dOh = dO + b * dos.sb + h * dos.sh;
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
@@ -846,16 +856,34 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
     }
   }
 
-  // ---- epilogue: write dK, dV (bf16)
-  unsigned short* dKh = dK + b * dks.sb + h * dks.sh;
-  unsigned short* dVh = dV + b * dvs.sb + h * dvs.sh;
+  // ---- epilogue: write dK, dV.  With GQA (several q heads per kv head)
+  // the partials accumulate into fp32 buffers with atomics; otherwise a
+  // direct bf16 store.
+  if (n_kv != n_heads) {
+    float* dKh = dKf + b * dks.sb + h_kv * dks.sh;
+    float* dVh = dVf + b * dvs.sb + h_kv * dvs.sh;
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    const int key_glob = kv0 + wid * 16 + qg * 4 + reg;
+    for (int reg = 0; reg < 4; ++reg) {
+      const int key_glob = kv0 + wid * 16 + qg * 4 + reg;
 #pragma unroll
-    for (int dt = 0; dt < D / 16; ++dt) {
-      dKh[(long)key_glob * dks.st + dt * 16 + r] = f2us(acc_dk[dt][reg]);
-      dVh[(long)key_glob * dvs.st + dt * 16 + r] = f2us(acc_dv[dt][reg]);
+      for (int dt = 0; dt < D / 16; ++dt) {
+        atomicAdd(&dKh[(long)key_glob * dks.st + dt * 16 + r],
+                  acc_dk[dt][reg]);
+        atomicAdd(&dVh[(long)key_glob * dvs.st + dt * 16 + r],
+                  acc_dv[dt][reg]);
+      }
+    }
+  } else {
+    unsigned short* dKh = dK + b * dks.sb + h * dks.sh;
+    unsigned short* dVh = dV + b * dvs.sb + h * dvs.sh;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int key_glob = kv0 + wid * 16 + qg * 4 + reg;
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        dKh[(long)key_glob * dks.st + dt * 16 + r] = f2us(acc_dk[dt][reg]);
+        dVh[(long)key_glob * dvs.st + dt * 16 + r] = f2us(acc_dv[dt][reg]);
+      }
     }
   }
 }
@@ -867,6 +895,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
   const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
             D = (int)q.size(3);
+  const int Hkv = (int)k.size(1);
+  TORCH_CHECK(H % Hkv == 0);
   TORCH_CHECK(T % KB == 0, "attn_bwd: T must be a multiple of 64");
   TORCH_CHECK(D == 64 || D == 128 || D == 256);
   TORCH_CHECK(dout.stride(3) == 1 && q.stride(3) == 1 && k.stride(3) == 1 &&
@@ -879,8 +909,16 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   // so the model-side transposes stay views)
   auto dq_f32 = at::zeros({B, T, H, D}, q.options().dtype(at::kFloat))
                     .permute({0, 2, 1, 3});
-  auto dk = at::empty({B, T, H, D}, k.options()).permute({0, 2, 1, 3});
-  auto dv = at::empty({B, T, H, D}, v.options()).permute({0, 2, 1, 3});
+  const bool gqa = Hkv != H;
+  auto dk = at::empty({B, T, Hkv, D}, k.options()).permute({0, 2, 1, 3});
+  auto dv = at::empty({B, T, Hkv, D}, v.options()).permute({0, 2, 1, 3});
+  at::Tensor dk_f32, dv_f32;
+  if (gqa) {
+    dk_f32 = at::zeros({B, T, Hkv, D}, q.options().dtype(at::kFloat))
+                 .permute({0, 2, 1, 3});
+    dv_f32 = at::zeros({B, T, Hkv, D}, q.options().dtype(at::kFloat))
+                 .permute({0, 2, 1, 3});
+  }
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
   dim3 grid(B * H, T / KB), block(256);
@@ -912,16 +950,21 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        lse.data_ptr<float>(), delta.data_ptr<float>(),       \
                        dq_f32.data_ptr<float>(),                             \
                        reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
-                       reinterpret_cast<unsigned short*>(dv.data_ptr()), T,  \
-                       H, scale, causal ? 1 : 0, str_of(dout), str_of(q),    \
-                       str_of(k), str_of(v), str_of(dq_f32), str_of(dk),     \
-                       str_of(dv));                                          \
+                       reinterpret_cast<unsigned short*>(dv.data_ptr()),     \
+                       gqa ? dk_f32.data_ptr<float>() : nullptr,             \
+                       gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,  \
+                       scale, causal ? 1 : 0, str_of(dout), str_of(q),       \
+                       str_of(k), str_of(v), str_of(dq_f32),                 \
+                       gqa ? str_of(dk_f32) : str_of(dk),                    \
+                       gqa ? str_of(dv_f32) : str_of(dv));                   \
   } while (0)
 
   if (D == 64) LAUNCH_B(64);
   else if (D == 128) LAUNCH_B(128);
   else LAUNCH_B(256);
 #undef LAUNCH_B
+  if (gqa) return {dq_f32.to(at::kBFloat16), dk_f32.to(at::kBFloat16),
+                   dv_f32.to(at::kBFloat16)};
   return {dq_f32.to(at::kBFloat16), dk, dv};
 }
 

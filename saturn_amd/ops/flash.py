@@ -79,10 +79,7 @@ def flash_attention(q, k, v, causal: bool = True) -> torch.Tensor:
     """q,k,v: [B, H, T, D] bf16/fp16 contiguous (GQA: H_kv may divide H)."""
     ext = require_ext()
     if hasattr(ext, "attn_fwd") and _kernel_supported(q):
-        if k.shape[1] != q.shape[1]:  # GQA: expand kv heads
-            rep = q.shape[1] // k.shape[1]
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
+        # GQA is native in the kernels: kv may keep fewer heads
         return _FlashFn.apply(
             _dense_rows(q), _dense_rows(k), _dense_rows(v), causal
         )

@@ -360,3 +360,40 @@ def test_full_attention_noncausal_vs_fp32():
     assert rel_err(q.grad, q32g.grad) < 5e-2
     assert rel_err(k.grad, k32g.grad) < 5e-2
     assert rel_err(v.grad, v32g.grad) < 5e-2
+
+
+def test_flash_attention_gqa_native():
+    """GQA without kv expansion: fwd+bwd vs the expanded fp32 reference;
+    dk/dv come back at the kv head count."""
+    from saturn_amd.ops import require_ext
+    from saturn_amd.ops.flash import flash_attention
+    from saturn_amd.ops.functional import attention_math
+
+    ext = require_ext()
+    if not hasattr(ext, "attn_fwd"):
+        pytest.skip("attn_fwd not built")
+    torch.manual_seed(0)
+    B, H, Hkv, T, D = 2, 8, 2, 128, 128
+    rep = H // Hkv
+    q32 = torch.randn(B, H, T, D, device="cuda")
+    k32 = torch.randn(B, Hkv, T, D, device="cuda")
+    v32 = torch.randn(B, Hkv, T, D, device="cuda")
+    q32.requires_grad_(True)
+    k32.requires_grad_(True)
+    v32.requires_grad_(True)
+    ref = attention_math(
+        q32, k32.repeat_interleave(rep, 1), v32.repeat_interleave(rep, 1)
+    )
+    do = torch.randn_like(ref)
+    ref.backward(do)
+
+    q = q32.detach().to(torch.bfloat16).requires_grad_(True)
+    k = k32.detach().to(torch.bfloat16).requires_grad_(True)
+    v = v32.detach().to(torch.bfloat16).requires_grad_(True)
+    out = flash_attention(q, k, v)
+    assert rel_err(out, ref) < 4e-2
+    out.backward(do.to(torch.bfloat16))
+    assert k.grad.shape == (B, Hkv, T, D)
+    assert rel_err(q.grad, q32.grad) < 5e-2
+    assert rel_err(k.grad, k32.grad) < 5e-2
+    assert rel_err(v.grad, v32.grad) < 5e-2
