@@ -17,6 +17,7 @@ from saturn_amd.ops.functional import (
     apply_rope,
     causal_attention,
     fused_cross_entropy,
+    fused_swiglu,
     rope_tables,
 )
 
@@ -82,9 +83,7 @@ class LlamaMLP(nn.Module):
         self.down_proj = nn.Linear(cfg.ffn_dim, cfg.n_embd, bias=False)
 
     def forward(self, x):
-        return self.down_proj(
-            torch.nn.functional.silu(self.gate_proj(x)) * self.up_proj(x)
-        )
+        return self.down_proj(fused_swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
 class LlamaBlock(nn.Module):

@@ -29,6 +29,9 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor o, at::Tensor lse,
                                  bool causal);
 at::Tensor add3(at::Tensor a, at::Tensor b, at::Tensor c);
+at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up);
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor gate,
+                                   at::Tensor up);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -43,4 +46,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
   m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward");
   m.def("add3", &samd::add3, "fused 3-way residual add");
+  m.def("swiglu_fwd", &samd::swiglu_fwd, "fused silu(gate)*up");
+  m.def("swiglu_bwd", &samd::swiglu_bwd, "fused SwiGLU backward");
 }

@@ -298,3 +298,27 @@ def fused_add3(a, b, c):
     if a.is_cuda:
         return _Add3Fn.apply(a, b, c)
     return a + b + c
+
+
+class _SwigluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        ext = require_ext()
+        gate = gate.contiguous()
+        up = up.contiguous()
+        ctx.save_for_backward(gate, up)
+        return ext.swiglu_fwd(gate, up)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = require_ext()
+        gate, up = ctx.saved_tensors
+        dgate, dup = ext.swiglu_bwd(dout.contiguous(), gate, up)
+        return dgate, dup
+
+
+def fused_swiglu(gate, up):
+    """silu(gate) * up in one pass each way (Llama FFN)."""
+    if gate.is_cuda:
+        return _SwigluFn.apply(gate, up)
+    return torch.nn.functional.silu(gate) * up

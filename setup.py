@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "rope.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "add3.hip"),
+        os.path.join(CSRC, "swiglu.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -397,3 +397,24 @@ def test_flash_attention_gqa_native():
     assert rel_err(q.grad, q32.grad) < 5e-2
     assert rel_err(k.grad, k32.grad) < 5e-2
     assert rel_err(v.grad, v32.grad) < 5e-2
+
+
+def test_swiglu_fwd_bwd():
+    ext = requires_ext()
+    torch.manual_seed(0)
+    n = 4096 * 3 + 5  # vector path + tail
+    g32 = torch.randn(n, device="cuda", requires_grad=True)
+    u32 = torch.randn(n, device="cuda", requires_grad=True)
+    ref = torch.nn.functional.silu(g32) * u32
+    do = torch.randn_like(ref)
+    ref.backward(do)
+
+    from saturn_amd.ops.functional import fused_swiglu
+
+    g = g32.detach().to(torch.bfloat16).requires_grad_(True)
+    u = u32.detach().to(torch.bfloat16).requires_grad_(True)
+    out = fused_swiglu(g, u)
+    out.backward(do.to(torch.bfloat16))
+    assert rel_err(out, ref) < 2e-2
+    assert rel_err(g.grad, g32.grad) < 3e-2
+    assert rel_err(u.grad, u32.grad) < 3e-2
