@@ -16,6 +16,7 @@ from saturn_amd.ops.functional import (
     FusedLayerNorm,
     full_attention,
     fused_cross_entropy,
+    FusedEmbedding,
 )
 
 
@@ -60,8 +61,8 @@ class BertForMaskedLM(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.cfg = cfg
-        self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
-        self.wpe = nn.Embedding(cfg.n_ctx, cfg.n_embd)
+        self.wte = FusedEmbedding(cfg.vocab_size, cfg.n_embd)
+        self.wpe = FusedEmbedding(cfg.n_ctx, cfg.n_embd)
         self.h = nn.ModuleList(BertBlock(cfg) for _ in range(cfg.n_layer))
         self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)

@@ -17,6 +17,7 @@ from saturn_amd.ops.functional import (
     FusedLayerNorm,
     causal_attention,
     fused_cross_entropy,
+    FusedEmbedding,
 )
 
 
@@ -78,8 +79,8 @@ class GPT2ForCausalLM(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
         self.cfg = cfg
-        self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
-        self.wpe = nn.Embedding(cfg.n_ctx, cfg.n_embd)
+        self.wte = FusedEmbedding(cfg.vocab_size, cfg.n_embd)
+        self.wpe = FusedEmbedding(cfg.n_ctx, cfg.n_embd)
         self.h = nn.ModuleList(GPT2Block(cfg) for _ in range(cfg.n_layer))
         self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)

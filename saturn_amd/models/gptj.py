@@ -30,6 +30,7 @@ from saturn_amd.ops.functional import (
     fused_add3,
     fused_cross_entropy,
     rope_tables,
+    FusedEmbedding,
 )
 
 
@@ -109,7 +110,7 @@ class GPTJForCausalLM(nn.Module):
     def __init__(self, cfg: GPTJConfig):
         super().__init__()
         self.cfg = cfg
-        self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
+        self.wte = FusedEmbedding(cfg.vocab_size, cfg.n_embd)
         self.h = nn.ModuleList(GPTJBlock(cfg) for _ in range(cfg.n_layer))
         self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=True)

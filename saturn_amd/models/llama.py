@@ -19,6 +19,7 @@ from saturn_amd.ops.functional import (
     fused_cross_entropy,
     fused_swiglu,
     rope_tables,
+    FusedEmbedding,
 )
 
 
@@ -104,7 +105,7 @@ class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         self.cfg = cfg
-        self.wte = nn.Embedding(cfg.vocab_size, cfg.n_embd)
+        self.wte = FusedEmbedding(cfg.vocab_size, cfg.n_embd)
         self.h = nn.ModuleList(LlamaBlock(cfg) for _ in range(cfg.n_layer))
         self.ln_f = FusedRMSNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)

@@ -32,6 +32,8 @@ at::Tensor add3(at::Tensor a, at::Tensor b, at::Tensor c);
 at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up);
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor gate,
                                    at::Tensor up);
+at::Tensor embed_fwd(at::Tensor weight, at::Tensor idx);
+at::Tensor embed_bwd(at::Tensor dout, at::Tensor idx, long vocab);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -48,4 +50,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add3", &samd::add3, "fused 3-way residual add");
   m.def("swiglu_fwd", &samd::swiglu_fwd, "fused silu(gate)*up");
   m.def("swiglu_bwd", &samd::swiglu_bwd, "fused SwiGLU backward");
+  m.def("embed_fwd", &samd::embed_fwd, "token embedding gather");
+  m.def("embed_bwd", &samd::embed_bwd, "embedding scatter-add backward");
 }

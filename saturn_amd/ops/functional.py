@@ -322,3 +322,47 @@ def fused_swiglu(gate, up):
     if gate.is_cuda:
         return _SwigluFn.apply(gate, up)
     return torch.nn.functional.silu(gate) * up
+
+
+class _EmbedFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, weight, idx):
+        ext = require_ext()
+        ctx.save_for_backward(idx)
+        ctx.vocab = weight.shape[0]
+        return ext.embed_fwd(weight, idx)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = require_ext()
+        (idx,) = ctx.saved_tensors
+        return ext.embed_bwd(dout, idx, ctx.vocab), None
+
+
+def fused_embedding(weight, idx):
+    """Token embedding gather; backward is a fused fp32-atomic scatter-add
+    into the vocab table (reference GPTJ.py:346,377)."""
+    if weight.is_cuda:
+        return _EmbedFn.apply(weight, idx)
+    return F.embedding(idx, weight)
+
+
+class FusedEmbedding(torch.nn.Embedding):
+    """Drop-in ``nn.Embedding`` that routes to the K5 HIP kernels on GPU.
+
+    Subclassing keeps state-dict keys, init-weight isinstance checks and TP
+    sharding untouched; exotic Embedding options (padding_idx, max_norm,
+    sparse) fall back to the stock op.
+    """
+
+    def forward(self, idx):  # noqa: D102
+        if (
+            self.weight.is_cuda
+            and self.padding_idx is None
+            and self.max_norm is None
+            and not self.sparse
+            and self.weight.dtype
+            in (torch.bfloat16, torch.float16, torch.float32)
+        ):
+            return _EmbedFn.apply(self.weight, idx)
+        return super().forward(idx)

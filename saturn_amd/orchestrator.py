@@ -12,6 +12,7 @@ from __future__ import annotations
 
 import logging
 from concurrent.futures import ThreadPoolExecutor
+from timeit import default_timer as timer
 from typing import List, Optional
 
 from saturn_amd.engine import execute, forecast
@@ -100,6 +101,7 @@ def orchestrate(
                 if next_tasks
                 else None
             )
+            t0 = timer()
             failed = execute(
                 relevant,
                 batches,
@@ -107,6 +109,15 @@ def orchestrate(
                 plan,
                 launch_timeout=launch_timeout,
                 raise_on_failure=False,
+            )
+            elapsed = timer() - t0
+            # interval accounting (reference executor.py:124-129): how far
+            # off the profiled batch times were from this interval's wall
+            log.info(
+                "interval wall %.1fs vs budget %.1fs (%+.1f%%)",
+                elapsed,
+                interval,
+                (elapsed - interval) / interval * 100.0,
             )
             if failed:
                 for t in failed:

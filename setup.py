@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "add3.hip"),
         os.path.join(CSRC, "swiglu.hip"),
+        os.path.join(CSRC, "embedding.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
