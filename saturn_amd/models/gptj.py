@@ -30,6 +30,7 @@ from saturn_amd.ops.functional import (
     fused_add3,
     fused_cross_entropy,
     rope_tables,
+    FusedDropout,
     FusedEmbedding,
 )
 
@@ -100,10 +101,12 @@ class GPTJBlock(nn.Module):
         self.ln_1 = FusedLayerNorm(cfg.n_embd)
         self.attn = GPTJAttention(cfg)
         self.mlp = GPTJMLP(cfg)
+        # reference resid_pdrop (GPTJ.py:95-96); 0.0 = exact passthrough
+        self.drop = FusedDropout(cfg.resid_pdrop)
 
     def forward(self, x):
         h = self.ln_1(x)
-        return fused_add3(x, self.attn(h), self.mlp(h))
+        return fused_add3(x, self.drop(self.attn(h)), self.drop(self.mlp(h)))
 
 
 class GPTJForCausalLM(nn.Module):
@@ -111,6 +114,7 @@ class GPTJForCausalLM(nn.Module):
         super().__init__()
         self.cfg = cfg
         self.wte = FusedEmbedding(cfg.vocab_size, cfg.n_embd)
+        self.drop = FusedDropout(cfg.embd_pdrop)
         self.h = nn.ModuleList(GPTJBlock(cfg) for _ in range(cfg.n_layer))
         self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=True)
@@ -126,7 +130,7 @@ class GPTJForCausalLM(nn.Module):
             nn.init.normal_(m.weight, std=0.02)
 
     def forward(self, input_ids):
-        x = self.wte(input_ids)
+        x = self.drop(self.wte(input_ids))
         for block in self.h:
             x = block(x)
         x = self.ln_f(x)

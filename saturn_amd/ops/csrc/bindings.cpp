@@ -34,6 +34,8 @@ std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor gate,
                                    at::Tensor up);
 at::Tensor embed_fwd(at::Tensor weight, at::Tensor idx);
 at::Tensor embed_bwd(at::Tensor dout, at::Tensor idx, long vocab);
+at::Tensor dropout_fwd(at::Tensor x, double p, long seed);
+at::Tensor dropout_bwd(at::Tensor dout, double p, long seed);
 }  // namespace samd
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -52,4 +54,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_bwd", &samd::swiglu_bwd, "fused SwiGLU backward");
   m.def("embed_fwd", &samd::embed_fwd, "token embedding gather");
   m.def("embed_bwd", &samd::embed_bwd, "embedding scatter-add backward");
+  m.def("dropout_fwd", &samd::dropout_fwd, "counter-based fused dropout");
+  m.def("dropout_bwd", &samd::dropout_bwd, "dropout backward (mask recompute)");
 }

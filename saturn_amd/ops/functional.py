@@ -366,3 +366,41 @@ class FusedEmbedding(torch.nn.Embedding):
         ):
             return _EmbedFn.apply(self.weight, idx)
         return super().forward(idx)
+
+
+class _DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed):
+        ext = require_ext()
+        ctx.p = p
+        ctx.seed = seed
+        return ext.dropout_fwd(x, p, seed)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = require_ext()
+        return ext.dropout_bwd(dout.contiguous(), ctx.p, ctx.seed), None, None
+
+
+def fused_dropout(x, p: float, training: bool = True):
+    """Counter-based dropout: the backward regenerates the keep
+    mask from a 64-bit seed instead of storing a mask tensor, so both passes
+    stay single-read/single-write HBM-bound (reference GPTJ.py:95-96,347)."""
+    if p == 0.0 or not training:
+        return x
+    if x.is_cuda and x.dtype in (torch.bfloat16, torch.float16):
+        seed = int(torch.randint(0, 2**62, (1,)).item())
+        return _DropoutFn.apply(x.contiguous(), p, seed)
+    return F.dropout(x, p, training)
+
+
+class FusedDropout(torch.nn.Module):
+    def __init__(self, p: float = 0.0):
+        super().__init__()
+        self.p = p
+
+    def forward(self, x):  # noqa: D102
+        return fused_dropout(x, self.p, self.training)
+
+    def extra_repr(self):  # noqa: D102
+        return f"p={self.p}"

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "add3.hip"),
         os.path.join(CSRC, "swiglu.hip"),
         os.path.join(CSRC, "embedding.hip"),
+        os.path.join(CSRC, "dropout.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

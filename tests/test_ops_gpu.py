@@ -456,3 +456,37 @@ def test_fused_embedding_module_matches_stock():
     out = e(idx)
     ref = torch.nn.functional.embedding(idx, e.weight)
     assert torch.equal(out, ref)
+
+
+def test_dropout_fwd_bwd_mask_consistency():
+    """K6: keep-rate ~ (1-p), kept values scaled by 1/(1-p), and the
+    backward regenerates exactly the forward's mask from the seed."""
+    requires_ext()
+    torch.manual_seed(3)
+    from saturn_amd.ops.functional import fused_dropout
+
+    n = 1 << 20
+    p = 0.3
+    x = torch.full((n,), 2.0, device="cuda", dtype=torch.bfloat16,
+                   requires_grad=True)
+    y = fused_dropout(x, p)
+    keep = (y != 0)
+    frac = keep.float().mean().item()
+    assert abs(frac - (1 - p)) < 5e-3, frac
+    # kept elements are x/(1-p) exactly (bf16 rounding of the scale only)
+    expect = torch.tensor(2.0 / (1 - p)).to(torch.bfloat16).cuda()
+    assert torch.all(y[keep] == expect)
+    g = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    y.backward(g)
+    # grad mask must be the SAME mask the forward drew
+    assert torch.all((x.grad != 0) == (keep & (g != 0)))
+
+
+def test_dropout_p0_and_eval_identity():
+    requires_ext()
+    from saturn_amd.ops.functional import FusedDropout, fused_dropout
+
+    x = torch.randn(4096, device="cuda", dtype=torch.bfloat16)
+    assert fused_dropout(x, 0.0) is x
+    d = FusedDropout(0.5).eval()
+    assert d(x) is x
