@@ -39,6 +39,11 @@ def main() -> None:
                     help="override layer count (default: model standard)")
     ap.add_argument("--model", type=str, default="gptj-6b")
     ap.add_argument("--bucket-mb", type=float, default=64.0)
+    ap.add_argument(
+        "--hipgraph",
+        action="store_true",
+        help="capture the step in a hipGraph and replay it (world 1 only)",
+    )
     args = ap.parse_args()
 
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
@@ -113,13 +118,27 @@ def main() -> None:
     g = torch.Generator(device="cpu").manual_seed(4321 + rank)
     x = torch.randint(0, vocab, (args.batch, args.seq), generator=g).to(device)
 
-    def step() -> None:
-        logits = ddp(x)
-        loss = loss_fn(logits, x)
-        loss.backward()
-        ddp.grad_sync()
-        opt.step()
-        ddp.zero_grad_buffers()
+    if args.hipgraph:
+        if world > 1:
+            raise SystemExit("--hipgraph is single-GPU only this round")
+        from saturn_amd.utils.graph_step import graphed_train_step
+
+        graphed, static_x = graphed_train_step(ddp.module, loss_fn, opt, x,
+                                               ddp=ddp)
+        static_x.copy_(x)
+
+        def step() -> None:
+            graphed.replay()
+
+    else:
+
+        def step() -> None:
+            logits = ddp(x)
+            loss = loss_fn(logits, x)
+            loss.backward()
+            ddp.grad_sync()
+            opt.step()
+            ddp.zero_grad_buffers()
 
     log(f"[rank {rank}] {n_params/1e9:.2f}B params; warmup {args.warmup} steps")
     for _ in range(args.warmup):

@@ -165,3 +165,56 @@ def test_bert_and_vit_tiny_train_gpu():
     loss = vit_loss(vit(px), y)
     loss.backward()
     assert torch.isfinite(torch.tensor(float(loss)))
+
+
+def test_hipgraph_step_matches_eager():
+    """A hipGraph-captured step must reproduce the eager param trajectory
+    bit-for-bit (same kernels, same order, static buffers)."""
+    import copy
+
+    from saturn_amd.models.gptj import get_gptj_model, pretraining_loss
+    from saturn_amd.ops.optim import FusedSGD
+    from saturn_amd.utils.graph_step import graphed_train_step
+
+    torch.manual_seed(0)
+    cfg = {"n_layer": 2, "n_embd": 256, "n_head": 4, "vocab_size": 512,
+           "n_ctx": 64, "rotary_dim": 16}
+    m1 = get_gptj_model(cfg).to("cuda", torch.bfloat16)
+    m2 = copy.deepcopy(m1)
+    o1 = FusedSGD(m1.parameters(), lr=1e-3)
+    o2 = FusedSGD(m2.parameters(), lr=1e-3)
+
+    g = torch.Generator().manual_seed(9)
+    batches = [torch.randint(0, 512, (2, 64), generator=g).cuda()
+               for _ in range(3)]
+
+    # eager trajectory (same zero/fwd/bwd/step order as the capture closure)
+    eager_losses = []
+    for b in batches:
+        o1.zero_grad(set_to_none=False)
+        loss = pretraining_loss(m1(b), b)
+        loss.backward()
+        o1.step()
+        eager_losses.append(float(loss))
+
+    # warmup mutates m2, so rebuild it identically after capture probing:
+    # instead, warm up AND capture on m2, then reset its state to m1's start
+    # is invalid — capture bakes pointers.  Correct scheme: warm up on a
+    # throwaway copy is impossible (pointers differ), so warm up on m2 and
+    # reset the *values* in place before replaying.
+    m3 = copy.deepcopy(m2)  # pristine values
+    graphed, static_x = graphed_train_step(
+        m2, pretraining_loss, o2, batches[0], warmup=2
+    )
+    with torch.no_grad():
+        for p2, p3 in zip(m2.parameters(), m3.parameters()):
+            p2.copy_(p3)
+    graph_losses = []
+    for b in batches:
+        static_x.copy_(b)
+        graph_losses.append(float(graphed.replay()))
+
+    assert eager_losses == pytest.approx(graph_losses, rel=0, abs=0), (
+        eager_losses, graph_losses)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.equal(p1, p2)
