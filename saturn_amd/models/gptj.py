@@ -183,6 +183,8 @@ def get_gptj_model(kwargs=None):
         n_ctx=kwargs.get("n_ctx", 512),
         vocab_size=kwargs.get("vocab_size", 50400),
         rotary_dim=kwargs.get("rotary_dim", 64),
+        embd_pdrop=kwargs.get("embd_pdrop", 0.0),
+        resid_pdrop=kwargs.get("resid_pdrop", 0.0),
     )
     torch.manual_seed(kwargs.get("seed", 0))
     return GPTJForCausalLM(cfg)
