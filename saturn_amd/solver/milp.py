@@ -314,7 +314,7 @@ def _milp_plan(task_list, n_gpus: int, timeout: float) -> Optional[Plan]:
         gpu_sets.append(
             [g for g in range(n_gpus) if round(xv[occ_off + t * n_gpus + g]) == 1]
         )
-    return Plan(
+    plan = Plan(
         task_names=[t.name for t in task_list],
         chosen_option=chosen,
         gpu_counts=[opts[t][chosen[t]][0] for t in range(T)],
@@ -323,6 +323,30 @@ def _milp_plan(task_list, n_gpus: int, timeout: float) -> Optional[Plan]:
         runtimes=[float(opts[t][chosen[t]][1]) for t in range(T)],
         makespan=float(xv[mk]),
         solver_status="optimal" if res.status == 0 else f"status_{res.status}",
+    )
+    _repair_overlaps(plan)
+    return plan
+
+
+def _repair_overlaps(plan: Plan) -> None:
+    """HiGHS honors constraints only to its ~1e-6 feasibility tolerance, so
+    two tasks sharing a GPU can overlap by a microsecond in the raw
+    solution.  Execution is dependency-ordered (engine events), so this only
+    tidies the reported schedule: push each start to the latest end among
+    earlier-started tasks sharing a GPU, in start order."""
+    order = sorted(range(len(plan.task_names)),
+                   key=lambda i: (plan.start_times[i], i))
+    for pos, i in enumerate(order):
+        floor = 0.0
+        for j in order[:pos]:
+            if set(plan.gpu_sets[i]) & set(plan.gpu_sets[j]):
+                floor = max(floor, plan.start_times[j] + plan.runtimes[j])
+        if plan.start_times[i] < floor:
+            plan.start_times[i] = floor
+    plan.makespan = max(
+        [plan.start_times[i] + plan.runtimes[i]
+         for i in range(len(plan.task_names))],
+        default=0.0,
     )
 
 

@@ -175,3 +175,65 @@ def test_solver_scales_to_larger_batches(save_dir):
     # must beat fully-sequential best-option schedule
     seq = sum(min(s.runtime for s in t.strategies.values()) for t in tasks)
     assert plan.makespan < seq
+
+
+# ---------------------------------------------------------------------------
+# Property-based: every solver output on random runtime tables is a valid
+# gang schedule (hypothesis shrinks violations to minimal cases)
+# ---------------------------------------------------------------------------
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@st.composite
+def _task_batch(draw):
+    n_gpus = draw(st.sampled_from([2, 4, 8]))
+    n_tasks = draw(st.integers(1, 5))
+    specs = []
+    for i in range(n_tasks):
+        counts = draw(
+            st.lists(
+                st.sampled_from([g for g in (1, 2, 4, 8) if g <= n_gpus]),
+                min_size=1, max_size=3, unique=True,
+            )
+        )
+        runtimes = {
+            g: draw(st.floats(0.5, 500.0, allow_nan=False)) for g in counts
+        }
+        specs.append(runtimes)
+    return n_gpus, specs
+
+
+@given(_task_batch())
+@settings(max_examples=25, deadline=None)
+def test_solver_plans_always_valid(batch):
+    import tempfile
+
+    n_gpus, specs = batch
+    with tempfile.TemporaryDirectory() as d:
+        tasks = [
+            make_task(f"t{i}", rts, d) for i, rts in enumerate(specs)
+        ]
+        plan = solve(tasks, n_gpus=n_gpus, timeout=5)
+        check_plan_valid(plan, tasks, n_gpus)
+        # greedy fallback obeys the same invariants
+        check_plan_valid(_greedy_plan(tasks, n_gpus), tasks, n_gpus)
+        # dependency_dict is acyclic and consistent with start order
+        deps = plan.dependency_dict()
+        for i, pre in deps.items():
+            for j in pre:
+                assert (plan.start_times[j], j) < (plan.start_times[i], i)
+                assert i not in deps[j], "dependency cycle"
+
+
+@given(st.floats(0.0, 1000.0), _task_batch())
+@settings(max_examples=10, deadline=None)
+def test_plan_shift_preserves_validity(shift_s, batch):
+    import tempfile
+
+    n_gpus, specs = batch
+    with tempfile.TemporaryDirectory() as d:
+        tasks = [make_task(f"t{i}", rts, d) for i, rts in enumerate(specs)]
+        plan = solve(tasks, n_gpus=n_gpus, timeout=5)
+        plan.shift(shift_s)
+        assert all(s >= 0.0 for s in plan.start_times)
+        assert plan.makespan >= 0.0
