@@ -678,7 +678,7 @@ __global__ void attn_delta_kernel(const unsigned short* __restrict__ dO,
   }
 }
 
-template <int D>
+template <int D, bool TR16>
 __launch_bounds__(256, 2)
 __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 const unsigned short* __restrict__ Q,
@@ -805,16 +805,22 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       bf16x8 ap = *reinterpret_cast<const bf16x8*>(&lds.pt[wid][r][qg * 8]);
       bf16x8 adst =
           *reinterpret_cast<const bf16x8*>(&lds.dst[wid * 16 + r][qg * 8]);
-      // NB: scalar gathers here beat tr16 reads — the asm's memory clobber
-      // would drain the freshly-issued pt/dst scalar writes at every step,
-      // while hipcc schedules these reads with counted waits (measured).
+      // Two B-fragment paths (host picks per SAMD_ATTN_BWD_TR16): scalar
+      // gathers let hipcc schedule with counted waits across the pt/dst
+      // stores; the tr16 hardware-transpose reads are denser but their asm
+      // clobber is a scheduling fence (measured both ways — see profiles/).
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         bf16x8 bdo, bq;
+        if constexpr (TR16) {
+          tr16_bfrag2<D + 8, D + 8>(&lds.dot[qg * 8][dt * 16],
+                                    &lds.qt[qg * 8][dt * 16], lane, bdo, bq);
+        } else {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          bdo[j] = (short)lds.dot[qg * 8 + j][dt * 16 + r];
-          bq[j] = (short)lds.qt[qg * 8 + j][dt * 16 + r];
+          for (int j = 0; j < 8; ++j) {
+            bdo[j] = (short)lds.dot[qg * 8 + j][dt * 16 + r];
+            bq[j] = (short)lds.qt[qg * 8 + j][dt * 16 + r];
+          }
         }
         acc_dv[dt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bdo, acc_dv[dt], 0, 0, 0);
@@ -836,10 +842,16 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
 #pragma unroll
         for (int kt2 = 0; kt2 < 2; ++kt2) {
           bf16x8 adst, bk;
+          if constexpr (TR16) {
+            tr16_bfrag2<bwd::QT + 8, D + 8>(
+                &lds.dst[kt2 * 32 + qg * 8][qsub * 16],
+                &lds.kt[kt2 * 32 + qg * 8][dt * 16], lane, adst, bk);
+          } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
-            bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
+            for (int j = 0; j < 8; ++j) {
+              adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
+              bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
+            }
           }
           // A[q row][key k]: dst is [key][q]; the transposed read above
           // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
@@ -922,6 +934,12 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   auto stream = at::hip::getCurrentHIPStream();
   const float scale = 1.f / sqrtf((float)D);
   dim3 grid(B * H, T / KB), block(256);
+  // default ON for D<=128 (measured +2-11% bwd across shapes; numerics
+  // validated both ways); SAMD_ATTN_BWD_TR16=0 reverts to scalar gathers
+  static const bool use_tr16 = [] {
+    const char* e = getenv("SAMD_ATTN_BWD_TR16");
+    return !(e && e[0] == '0');
+  }();
 
 #define LAUNCH_B(DD)                                                         \
   do {                                                                       \
@@ -935,13 +953,35 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     size_t shmem = sizeof(AttnBwdLds<DD>);                                   \
     static bool attr_set_##DD = [] {                                         \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&attn_bwd_kernel<DD>),               \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, false>),        \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
+          (int)sizeof(AttnBwdLds<DD>));                                      \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, true>),         \
           hipFuncAttributeMaxDynamicSharedMemorySize,                        \
           (int)sizeof(AttnBwdLds<DD>));                                      \
       return true;                                                           \
     }();                                                                     \
     (void)attr_set_##DD;                                                     \
-    hipLaunchKernelGGL((attn_bwd_kernel<DD>), grid, block, shmem,            \
+    if (use_tr16 && DD <= 128)                                               \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, true>), grid, block, shmem,    \
+                       stream.stream(),                                      \
+                       reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
+                       reinterpret_cast<const unsigned short*>(q.data_ptr()),  \
+                       reinterpret_cast<const unsigned short*>(k.data_ptr()),  \
+                       reinterpret_cast<const unsigned short*>(v.data_ptr()),  \
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),       \
+                       dq_f32.data_ptr<float>(),                             \
+                       reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
+                       reinterpret_cast<unsigned short*>(dv.data_ptr()),     \
+                       gqa ? dk_f32.data_ptr<float>() : nullptr,             \
+                       gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,  \
+                       scale, causal ? 1 : 0, str_of(dout), str_of(q),       \
+                       str_of(k), str_of(v), str_of(dq_f32),                 \
+                       gqa ? str_of(dk_f32) : str_of(dk),                    \
+                       gqa ? str_of(dv_f32) : str_of(dv));                   \
+    else                                                                     \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, false>), grid, block, shmem,   \
                        stream.stream(),                                      \
                        reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
                        reinterpret_cast<const unsigned short*>(q.data_ptr()),  \
