@@ -623,10 +623,10 @@ constexpr int QT = 32;   // q rows per tile
 // global 16-B vectors — measured +39% at the GPT-J backward shape.  At
 // D<=128 the struct fits two blocks/CU WITH the V tile, and the LDS copy
 // is faster (keeping it avoided a measured regression at T=2048/D=128).
-template <int D>
+template <int D, bool HASV>
 struct AttnBwdLds {
   unsigned short kt[bwd::KB][D + 8];
-  unsigned short vt[D <= 128 ? bwd::KB : 1][D + 8];
+  unsigned short vt[HASV ? bwd::KB : 1][D + 8];
   unsigned short qt[bwd::QT][D + 8];
   unsigned short dot[bwd::QT][D + 8];
   unsigned short pt[4][16][bwd::QT + 8];   // per-wave P^T tile
@@ -678,7 +678,7 @@ __global__ void attn_delta_kernel(const unsigned short* __restrict__ dO,
   }
 }
 
-template <int D, bool TR16>
+template <int D, bool TR16, bool VL2>
 __launch_bounds__(256, 2)
 __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 const unsigned short* __restrict__ Q,
@@ -696,7 +696,8 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 TStr dks, TStr dvs) {
   using namespace bwd;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  AttnBwdLds<D>& lds = *reinterpret_cast<AttnBwdLds<D>*>(smem);
+  constexpr bool HASV = (D <= 128) && !VL2;  // V tile LDS-resident?
+  AttnBwdLds<D, HASV>& lds = *reinterpret_cast<AttnBwdLds<D, HASV>*>(smem);
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -724,7 +725,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
           *reinterpret_cast<const bf16x8*>(
               Kh + (long)(kv0 + row) * ks.st + col);
-      if constexpr (D <= 128)
+      if constexpr (HASV)
         *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
             *reinterpret_cast<const bf16x8*>(
                 Vh + (long)(kv0 + row) * vs.st + col);
@@ -772,7 +773,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         bf16x8 ak = *reinterpret_cast<const bf16x8*>(
             &lds.kt[wid * 16 + r][ds * 32 + qg * 8]);
         bf16x8 av;
-        if constexpr (D <= 128)
+        if constexpr (HASV)
           av = *reinterpret_cast<const bf16x8*>(
               &lds.vt[wid * 16 + r][ds * 32 + qg * 8]);
         else
@@ -940,6 +941,29 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     const char* e = getenv("SAMD_ATTN_BWD_TR16");
     return !(e && e[0] == '0');
   }();
+  // V tile from L2 instead of LDS at D<=128 (SAMD_ATTN_BWD_VL2=1): trades
+  // per-q-tile L2 re-reads for 62->45 KB LDS/block (3 blocks/CU scalar path)
+  static const bool use_vl2 = [] {
+    const char* e = getenv("SAMD_ATTN_BWD_VL2");
+    return e && e[0] == '1';
+  }();
+
+#define BWD_ARGS                                                             \
+    stream.stream(),                                                         \
+    reinterpret_cast<const unsigned short*>(dout.data_ptr()),                \
+    reinterpret_cast<const unsigned short*>(q.data_ptr()),                   \
+    reinterpret_cast<const unsigned short*>(k.data_ptr()),                   \
+    reinterpret_cast<const unsigned short*>(v.data_ptr()),                   \
+    lse.data_ptr<float>(), delta.data_ptr<float>(),                          \
+    dq_f32.data_ptr<float>(),                                                \
+    reinterpret_cast<unsigned short*>(dk.data_ptr()),                        \
+    reinterpret_cast<unsigned short*>(dv.data_ptr()),                        \
+    gqa ? dk_f32.data_ptr<float>() : nullptr,                                \
+    gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,                     \
+    scale, causal ? 1 : 0, str_of(dout), str_of(q),                          \
+    str_of(k), str_of(v), str_of(dq_f32),                                    \
+    gqa ? str_of(dk_f32) : str_of(dk),                                       \
+    gqa ? str_of(dv_f32) : str_of(dv)
 
 #define LAUNCH_B(DD)                                                         \
   do {                                                                       \
@@ -950,53 +974,39 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        reinterpret_cast<const unsigned short*>(o.data_ptr()), \
                        delta.data_ptr<float>(), T, H, str_of(dout),          \
                        str_of(o));                                           \
-    size_t shmem = sizeof(AttnBwdLds<DD>);                                   \
+    const bool tr = use_tr16 && DD <= 128;                                   \
+    const bool vl2 = use_vl2 && DD <= 128;                                   \
+    constexpr size_t SH_V = sizeof(AttnBwdLds<DD, (DD <= 128)>);             \
+    constexpr size_t SH_N = sizeof(AttnBwdLds<DD, false>);                   \
+    size_t shmem = vl2 ? SH_N : SH_V;                                        \
     static bool attr_set_##DD = [] {                                         \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, false>),        \
-          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
-          (int)sizeof(AttnBwdLds<DD>));                                      \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, false, false>), \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_V);            \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, true>),         \
-          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
-          (int)sizeof(AttnBwdLds<DD>));                                      \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, true, false>),  \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_V);            \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, false, true>),  \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_N);            \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_bwd_kernel<DD, true, true>),   \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_N);            \
       return true;                                                           \
     }();                                                                     \
     (void)attr_set_##DD;                                                     \
-    if (use_tr16 && DD <= 128)                                               \
-      hipLaunchKernelGGL((attn_bwd_kernel<DD, true>), grid, block, shmem,    \
-                       stream.stream(),                                      \
-                       reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
-                       reinterpret_cast<const unsigned short*>(q.data_ptr()),  \
-                       reinterpret_cast<const unsigned short*>(k.data_ptr()),  \
-                       reinterpret_cast<const unsigned short*>(v.data_ptr()),  \
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),       \
-                       dq_f32.data_ptr<float>(),                             \
-                       reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
-                       reinterpret_cast<unsigned short*>(dv.data_ptr()),     \
-                       gqa ? dk_f32.data_ptr<float>() : nullptr,             \
-                       gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,  \
-                       scale, causal ? 1 : 0, str_of(dout), str_of(q),       \
-                       str_of(k), str_of(v), str_of(dq_f32),                 \
-                       gqa ? str_of(dk_f32) : str_of(dk),                    \
-                       gqa ? str_of(dv_f32) : str_of(dv));                   \
+    if (tr && vl2)                                                           \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, true, true>), grid, block,     \
+                         shmem, BWD_ARGS);                                   \
+    else if (tr)                                                             \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, true, false>), grid, block,    \
+                         shmem, BWD_ARGS);                                   \
+    else if (vl2)                                                            \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, false, true>), grid, block,    \
+                         shmem, BWD_ARGS);                                   \
     else                                                                     \
-      hipLaunchKernelGGL((attn_bwd_kernel<DD, false>), grid, block, shmem,   \
-                       stream.stream(),                                      \
-                       reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
-                       reinterpret_cast<const unsigned short*>(q.data_ptr()),  \
-                       reinterpret_cast<const unsigned short*>(k.data_ptr()),  \
-                       reinterpret_cast<const unsigned short*>(v.data_ptr()),  \
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),       \
-                       dq_f32.data_ptr<float>(),                             \
-                       reinterpret_cast<unsigned short*>(dk.data_ptr()),     \
-                       reinterpret_cast<unsigned short*>(dv.data_ptr()),     \
-                       gqa ? dk_f32.data_ptr<float>() : nullptr,             \
-                       gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,  \
-                       scale, causal ? 1 : 0, str_of(dout), str_of(q),       \
-                       str_of(k), str_of(v), str_of(dq_f32),                 \
-                       gqa ? str_of(dk_f32) : str_of(dk),                    \
-                       gqa ? str_of(dv_f32) : str_of(dv));                   \
+      hipLaunchKernelGGL((attn_bwd_kernel<DD, false, false>), grid, block,   \
+                         shmem, BWD_ARGS);                                   \
   } while (0)
 
   if (D == 64) LAUNCH_B(64);
