@@ -45,6 +45,23 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
         model = tp_shard_model(model)
         optimizer = _make_optimizer(task, model)
 
+        # Per-rank optimizer shard resume (mirrors FSDPExecutor: the
+        # reference loses moments at every interval; TP shards
+        # are rank-local so each rank reloads its own state when the solver
+        # kept the same world size)
+        import os as _os
+
+        opt_path = _os.path.join(
+            task.save_dir, f"{task.name}.tpopt.w{world}.r{rank}.pt"
+        )
+        if not trial and _os.path.isfile(opt_path):
+            try:
+                optimizer.load_state_dict(
+                    torch.load(opt_path, map_location="cpu", weights_only=False)
+                )
+            except Exception:
+                pass  # layout changed; fresh moments
+
         it = task.get_iterator() if not trial else task.get_fresh_iterator()
 
         def next_batch():
@@ -83,6 +100,9 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
         sd = tp_full_state_dict(model)
         if rank == 0 and sd is not None:
             task.save_checkpoint(sd, None)
+        tmp = opt_path + ".tmp"
+        torch.save(optimizer.state_dict(), tmp)
+        _os.replace(tmp, opt_path)
         if world > 1:
             dist.barrier()
         return None

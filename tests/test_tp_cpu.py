@@ -87,3 +87,35 @@ def test_megatron_rejects_single_gpu(save_dir):
     )
     params, _ = MegatronExecutor.search(t, [0], 932)
     assert params is None
+
+
+def test_tp_optimizer_shard_resume(save_dir, library_path):
+    """Second interval reloads each rank's optimizer shard (Adam moments)."""
+    import os
+
+    import torch
+
+    from saturn_amd import HParams, Strategy, Task
+    from saturn_amd.executors.megatron import MegatronExecutor
+    from saturn_amd.models import get_mlp_dataloader, get_mlp_model, mse_loss
+
+    t = Task(
+        get_mlp_model,
+        get_mlp_dataloader,
+        mse_loss,
+        HParams(lr=1e-2, batch_count=6, optimizer_cls=torch.optim.Adam),
+        gpu_range=[2],
+        name="tp_opt",
+        save_dir=save_dir,
+    )
+    t.strategies[2] = Strategy(MegatronExecutor, 2, {"tp": 2}, 6.0,
+                               batch_time=1.0)
+    t.select_strategy(t.strategies[2])
+    MegatronExecutor.execute(t, [0, 1], 931, 3)
+    for r in (0, 1):
+        shard = os.path.join(save_dir, f"tp_opt.tpopt.w2.r{r}.pt")
+        assert os.path.isfile(shard), shard
+        st = torch.load(shard, weights_only=False)
+        assert any("exp_avg" in v for v in st["state"].values())
+    MegatronExecutor.execute(t, [0, 1], 931, 3)  # second interval loads back
+    assert t.has_ckpt()
