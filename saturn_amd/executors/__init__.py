@@ -4,5 +4,6 @@ from .pipeline import PipelineExecutor
 from .megatron import MegatronExecutor
 from .spilled import SpilledExecutor
 from .ulysses import UlyssesExecutor
+from .expert import ExpertParallelExecutor
 
-__all__ = ["DDPExecutor", "FSDPExecutor", "PipelineExecutor", "SpilledExecutor", "MegatronExecutor", "UlyssesExecutor"]
+__all__ = ["DDPExecutor", "FSDPExecutor", "PipelineExecutor", "SpilledExecutor", "MegatronExecutor", "UlyssesExecutor", "ExpertParallelExecutor"]

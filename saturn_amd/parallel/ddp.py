@@ -62,10 +62,15 @@ class BucketedDDP(torch.nn.Module):
         process_group=None,
         grad_dtype: Optional[torch.dtype] = None,
         comm=None,
+        exclude=None,
     ) -> None:
         """``comm``: an optional saturn_amd native RcclComm; when given,
         bucket all-reduces run on its dedicated HIP stream (ncclAvg)
-        instead of torch.distributed's process group."""
+        instead of torch.distributed's process group.
+
+        ``exclude``: parameters to leave out of broadcast + bucketed
+        all-reduce (expert-parallel shards own their grads; see
+        parallel/expert.py)."""
         super().__init__()
         self.module = module
         self.pg = process_group
@@ -79,17 +84,23 @@ class BucketedDDP(torch.nn.Module):
 
         # Broadcast initial parameters from rank 0 (reference relies on the
         # DDP ctor for this, DDP.py:90).
+        skip = {id(p) for p in (exclude or ())}
         if self.world > 1:
             with torch.no_grad():
                 if comm is not None:
                     for p in module.parameters():
-                        comm.broadcast(p.data, 0)
+                        if id(p) not in skip:
+                            comm.broadcast(p.data, 0)
                     comm.join()
                 else:
                     for p in module.parameters():
-                        dist.broadcast(p.data, src=0, group=self.pg)
+                        if id(p) not in skip:
+                            dist.broadcast(p.data, src=0, group=self.pg)
 
-        params = [p for p in module.parameters() if p.requires_grad]
+        params = [
+            p for p in module.parameters()
+            if p.requires_grad and id(p) not in skip
+        ]
         bucket_bytes = int(bucket_mb * 1024 * 1024)
         self.buckets: List[_Bucket] = []
         self._param_bucket = {}
