@@ -237,3 +237,22 @@ def test_plan_shift_preserves_validity(shift_s, batch):
         plan.shift(shift_s)
         assert all(s >= 0.0 for s in plan.start_times)
         assert plan.makespan >= 0.0
+
+
+def test_solver_scales_to_wide_batch(save_dir):
+    """12 tasks x 3 options on 8 GPUs: the MILP (or its greedy fallback)
+    must return a valid plan within the timeout — the reference's largest
+    quoted batches are this order (BASELINE configs 3-4)."""
+    import random
+
+    rng = random.Random(7)
+    tasks = []
+    for i in range(12):
+        rts = {g: rng.uniform(50, 400) / g for g in (1, 2, 4)}
+        tasks.append(make_task(f"w{i}", rts, save_dir))
+    import time
+
+    t0 = time.monotonic()
+    plan = solve(tasks, n_gpus=8, timeout=20)
+    assert time.monotonic() - t0 < 60
+    check_plan_valid(plan, tasks, 8)
