@@ -178,3 +178,27 @@ def test_orchestrate_gives_up_after_max_retries(save_dir, library_path):
     with pytest.raises(RuntimeError, match="giving up"):
         orchestrate([t], interval=1e6, n_gpus=1, solver_timeout=5,
                     max_task_retries=1)
+
+
+def test_orchestrate_four_gpu_gang(save_dir, library_path):
+    """A 4-process gloo gang through the full pipeline (the driver's scale
+    bench runs world 4/8 on hardware; this pins the rendezvous + gang
+    binding shape beyond world 2)."""
+    from saturn_amd import HParams, Strategy, Task, orchestrate
+    from saturn_amd.executors.ddp import DDPExecutor
+    from saturn_amd.models import get_mlp_dataloader, get_mlp_model, mse_loss
+
+    t = Task(
+        get_mlp_model,
+        get_mlp_dataloader,
+        mse_loss,
+        HParams(lr=1e-2, batch_count=4),
+        gpu_range=[4],
+        name="gang4",
+        save_dir=save_dir,
+    )
+    t.strategies[4] = Strategy(DDPExecutor, 4, {"bucket_mb": 32.0}, 4.0,
+                               batch_time=1.0)
+    t.select_strategy(t.strategies[4])
+    orchestrate([t], interval=120, n_gpus=4)
+    assert t.has_ckpt()
