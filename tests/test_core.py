@@ -153,3 +153,34 @@ def test_text_dataloader_cache_roundtrip(tmp_path):
     dl = make_text_dataloader(str(p), batch_size=4, context_length=64)()
     x, y = next(iter(dl))
     assert x.shape == (4, 64) and torch.equal(x, y)
+
+
+def test_hparams_validation_errors():
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        HParams(lr=1e-3)  # neither epochs nor batch_count
+    with _pytest.raises(Exception):
+        HParams(lr=1e-3, epochs=1, batch_count=5)  # both
+
+
+def test_strategy_validation():
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        Strategy(None, 0)  # gpu count must be > 0
+    with _pytest.raises(ValueError):
+        Strategy(None, -2)
+    # feasible requires a real executor AND params (the reference's DDP
+    # was never selectable because it returned params=None — quirk #2)
+    s = Strategy(None, 2, runtime=10.0, batch_time=1.0)
+    assert not s.feasible
+    s2 = Strategy(object, 2, {"p": 1}, runtime=10.0, batch_time=1.0)
+    assert s2.feasible
+
+
+def test_solve_empty_task_list():
+    from saturn_amd.solver import solve
+
+    plan = solve([], n_gpus=4, timeout=5)
+    assert plan.task_names == [] and plan.makespan == 0.0
