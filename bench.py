@@ -121,6 +121,8 @@ def main() -> None:
     if args.hipgraph:
         if world > 1:
             raise SystemExit("--hipgraph is single-GPU only this round")
+        if not use_gpu:
+            raise SystemExit("--hipgraph needs a GPU (hipGraph capture)")
         from saturn_amd.utils.graph_step import graphed_train_step
 
         graphed, static_x = graphed_train_step(ddp.module, loss_fn, opt, x,
