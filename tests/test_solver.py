@@ -256,3 +256,27 @@ def test_solver_scales_to_wide_batch(save_dir):
     plan = solve(tasks, n_gpus=8, timeout=20)
     assert time.monotonic() - t0 < 60
     check_plan_valid(plan, tasks, 8)
+
+
+def test_convert_into_comprehensible_and_restrict(save_dir):
+    """The reference-parity plan conversion (milp.py:448-513) and the
+    introspection restrict path."""
+    from saturn_amd.solver import convert_into_comprehensible
+
+    a = make_task("a", {1: 30.0}, save_dir)
+    b = make_task("b", {1: 20.0}, save_dir)
+    plan = solve([a, b], n_gpus=1, timeout=10)  # forced serial on 1 GPU
+    nodes, deps, starts = convert_into_comprehensible([a, b], plan)
+    assert set(nodes.values()) == {0}
+    # exactly one of the two tasks depends on the other (shared GPU)
+    dep_counts = sorted(len(v) for v in deps.values())
+    assert dep_counts == [0, 1]
+    assert len(starts) == 2
+    # tasks got their strategies applied
+    assert a.selected_strategy is not None
+    assert b.selected_strategy is not None
+
+    # restrict to the surviving task keeps its scheduling entry
+    sub = plan.restrict(["b"])
+    assert sub.task_names == ["b"]
+    assert len(sub.start_times) == 1 and len(sub.gpu_sets) == 1
