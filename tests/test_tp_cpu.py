@@ -41,6 +41,20 @@ def _tp2_worker(rank, world, _):
         g_tp = m.wte.weight.grad
         g_ref = ref.wte.weight.grad
         assert torch.allclose(g_tp, g_ref, atol=1e-5), "replicated grad mismatch"
+        # sharded grads equal the dense model's corresponding slices:
+        # column-parallel q_proj -> rows, row-parallel out_proj -> cols
+        out_f = ref.h[0].attn.q_proj.weight.shape[0]
+        sl = slice(rank * out_f // world, (rank + 1) * out_f // world)
+        assert torch.allclose(
+            m.h[0].attn.q_proj.weight.grad,
+            ref.h[0].attn.q_proj.weight.grad[sl], atol=1e-5,
+        ), "column-parallel shard grad mismatch"
+        in_f = ref.h[0].attn.out_proj.weight.shape[1]
+        slc = slice(rank * in_f // world, (rank + 1) * in_f // world)
+        assert torch.allclose(
+            m.h[0].attn.out_proj.weight.grad,
+            ref.h[0].attn.out_proj.weight.grad[:, slc], atol=1e-5,
+        ), "row-parallel shard grad mismatch"
         sd = tp_full_state_dict(m)
         if rank == 0:
             ref_sd = ref.state_dict()
