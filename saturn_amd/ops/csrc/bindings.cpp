@@ -5,12 +5,13 @@
 
 namespace samd {
 void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
-               std::vector<at::Tensor> moms, double lr, double momentum,
-               double weight_decay);
+               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
+               double lr, double momentum, double weight_decay);
 void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double beta1, double beta2, double eps,
-                double weight_decay, double bc1, double bc2);
+                std::vector<at::Tensor> masters, double lr, double beta1,
+                double beta2, double eps, double weight_decay, double bc1,
+                double bc2);
 std::vector<at::Tensor> norm_fwd(at::Tensor x, at::Tensor w,
                                  c10::optional<at::Tensor> b, double eps,
                                  bool rms);

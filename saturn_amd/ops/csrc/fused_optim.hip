@@ -28,16 +28,18 @@ struct SgdArgs {
   void* p[MAX_T];
   void* g[MAX_T];
   void* m[MAX_T];       // momentum buffers (fp32; nullptr set if unused)
+  void* w[MAX_T];       // fp32 master weights (nullptr set if unused)
   long cum[MAX_T + 1];  // cumulative GROUP counts
   long numel[MAX_T];
   int n_tensors;
   float lr, momentum, weight_decay;
 };
 
-template <typename T, bool HAS_M>
-__device__ __forceinline__ void sgd_elem(T* p, const T* g, float* m, long j,
-                                         const SgdArgs& a) {
-  float pv = toF<T>(p[j]);
+template <typename T, bool HAS_M, bool HAS_W>
+__device__ __forceinline__ void sgd_elem(T* p, const T* g, float* m, float* w,
+                                         long j, const SgdArgs& a) {
+  // HAS_W: the fp32 master is the source of truth; p gets the rounded copy.
+  float pv = HAS_W ? w[j] : toF<T>(p[j]);
   float gv = toF<T>(g[j]);
   if (a.weight_decay != 0.f) gv += a.weight_decay * pv;
   if (HAS_M) {
@@ -45,10 +47,12 @@ __device__ __forceinline__ void sgd_elem(T* p, const T* g, float* m, long j,
     m[j] = mv;
     gv = mv;
   }
-  p[j] = fromF<T>(pv - a.lr * gv);
+  float nv = pv - a.lr * gv;
+  if (HAS_W) w[j] = nv;
+  p[j] = fromF<T>(nv);
 }
 
-template <typename T, bool HAS_M>
+template <typename T, bool HAS_M, bool HAS_W>
 __global__ void fused_sgd_kernel(SgdArgs a) {
   const long total = a.cum[a.n_tensors];  // total groups
   const long stride = (long)gridDim.x * blockDim.x;
@@ -63,23 +67,28 @@ __global__ void fused_sgd_kernel(SgdArgs a) {
     T* p = reinterpret_cast<T*>(a.p[t]);
     const T* g = reinterpret_cast<const T*>(a.g[t]);
     float* m = HAS_M ? reinterpret_cast<float*>(a.m[t]) : nullptr;
+    float* w = HAS_W ? reinterpret_cast<float*>(a.w[t]) : nullptr;
     const long n = a.numel[t];
     if constexpr (sizeof(T) == 2) {
       if (base + GRP <= n) {
         // vector path: 8 x 16-bit in one 16-B load/store
         short8v_ pv = *reinterpret_cast<short8v_*>(p + base);
         short8v_ gv = *reinterpret_cast<const short8v_*>(g + base);
-        float4v_ m0, m1;
+        float4v_ m0, m1, w0, w1;
         if (HAS_M) {
           m0 = *reinterpret_cast<float4v_*>(m + base);
           m1 = *reinterpret_cast<float4v_*>(m + base + 4);
+        }
+        if (HAS_W) {
+          w0 = *reinterpret_cast<float4v_*>(w + base);
+          w1 = *reinterpret_cast<float4v_*>(w + base + 4);
         }
 #pragma unroll
         for (int e = 0; e < GRP; ++e) {
           T pe, ge;
           pe.x = (unsigned short)pv[e];
           ge.x = (unsigned short)gv[e];
-          float pf = toF<T>(pe);
+          float pf = HAS_W ? (e < 4 ? w0[e] : w1[e - 4]) : toF<T>(pe);
           float gf = toF<T>(ge);
           if (a.weight_decay != 0.f) gf += a.weight_decay * pf;
           if (HAS_M) {
@@ -87,7 +96,11 @@ __global__ void fused_sgd_kernel(SgdArgs a) {
             if (e < 4) m0[e] = mv; else m1[e - 4] = mv;
             gf = mv;
           }
-          T out = fromF<T>(pf - a.lr * gf);
+          float nf = pf - a.lr * gf;
+          if (HAS_W) {
+            if (e < 4) w0[e] = nf; else w1[e - 4] = nf;
+          }
+          T out = fromF<T>(nf);
           pv[e] = (short)out.x;
         }
         *reinterpret_cast<short8v_*>(p + base) = pv;
@@ -95,12 +108,18 @@ __global__ void fused_sgd_kernel(SgdArgs a) {
           *reinterpret_cast<float4v_*>(m + base) = m0;
           *reinterpret_cast<float4v_*>(m + base + 4) = m1;
         }
+        if (HAS_W) {
+          *reinterpret_cast<float4v_*>(w + base) = w0;
+          *reinterpret_cast<float4v_*>(w + base + 4) = w1;
+        }
       } else {
-        for (long j = base; j < n; ++j) sgd_elem<T, HAS_M>(p, g, m, j, a);
+        for (long j = base; j < n; ++j)
+          sgd_elem<T, HAS_M, HAS_W>(p, g, m, w, j, a);
       }
     } else {
       const long end = (base + GRP <= n) ? base + GRP : n;
-      for (long j = base; j < end; ++j) sgd_elem<T, HAS_M>(p, g, m, j, a);
+      for (long j = base; j < end; ++j)
+        sgd_elem<T, HAS_M, HAS_W>(p, g, m, w, j, a);
     }
   }
 }
@@ -110,15 +129,16 @@ struct AdamArgs {
   void* g[MAX_T];
   void* m[MAX_T];
   void* v[MAX_T];
+  void* w[MAX_T];  // fp32 master weights (nullptr set if unused)
   long cum[MAX_T + 1];
   long numel[MAX_T];
   int n_tensors;
   float lr, beta1, beta2, eps, weight_decay, bc1, bc2;
 };
 
-template <typename T>
+template <typename T, bool HAS_W>
 __device__ __forceinline__ void adam_elem(T* p, const T* g, float* m, float* v,
-                                          long j, const AdamArgs& a,
+                                          float* w, long j, const AdamArgs& a,
                                           float inv_bc1, float inv_bc2) {
   float gv = toF<T>(g[j]);
   float mv = a.beta1 * m[j] + (1.f - a.beta1) * gv;
@@ -127,12 +147,14 @@ __device__ __forceinline__ void adam_elem(T* p, const T* g, float* m, float* v,
   v[j] = vv;
   float denom = sqrtf(vv * inv_bc2) + a.eps;
   float upd = (mv * inv_bc1) / denom;
-  float pv = toF<T>(p[j]);
+  float pv = HAS_W ? w[j] : toF<T>(p[j]);
   if (a.weight_decay != 0.f) pv *= (1.f - a.lr * a.weight_decay);
-  p[j] = fromF<T>(pv - a.lr * upd);
+  float nv = pv - a.lr * upd;
+  if (HAS_W) w[j] = nv;
+  p[j] = fromF<T>(nv);
 }
 
-template <typename T>
+template <typename T, bool HAS_W>
 __global__ void fused_adam_kernel(AdamArgs a) {
   const long total = a.cum[a.n_tensors];
   const long stride = (long)gridDim.x * blockDim.x;
@@ -150,6 +172,7 @@ __global__ void fused_adam_kernel(AdamArgs a) {
     const T* g = reinterpret_cast<const T*>(a.g[t]);
     float* m = reinterpret_cast<float*>(a.m[t]);
     float* v = reinterpret_cast<float*>(a.v[t]);
+    float* w = HAS_W ? reinterpret_cast<float*>(a.w[t]) : nullptr;
     const long n = a.numel[t];
     if constexpr (sizeof(T) == 2) {
       if (base + GRP <= n) {
@@ -160,6 +183,11 @@ __global__ void fused_adam_kernel(AdamArgs a) {
         float4v_ m1 = *reinterpret_cast<float4v_*>(m + base + 4);
         float4v_ v0 = *reinterpret_cast<float4v_*>(v + base);
         float4v_ v1 = *reinterpret_cast<float4v_*>(v + base + 4);
+        float4v_ w0, w1;
+        if (HAS_W) {
+          w0 = *reinterpret_cast<float4v_*>(w + base);
+          w1 = *reinterpret_cast<float4v_*>(w + base + 4);
+        }
 #pragma unroll
         for (int e = 0; e < GRP; ++e) {
           T ge;
@@ -172,9 +200,13 @@ __global__ void fused_adam_kernel(AdamArgs a) {
           float upd = (mv * inv_bc1) / denom;
           T pe;
           pe.x = (unsigned short)pv[e];
-          float pf = toF<T>(pe);
+          float pf = HAS_W ? (e < 4 ? w0[e] : w1[e - 4]) : toF<T>(pe);
           if (a.weight_decay != 0.f) pf *= (1.f - a.lr * a.weight_decay);
-          T out = fromF<T>(pf - a.lr * upd);
+          float nf = pf - a.lr * upd;
+          if (HAS_W) {
+            if (e < 4) w0[e] = nf; else w1[e - 4] = nf;
+          }
+          T out = fromF<T>(nf);
           pv[e] = (short)out.x;
         }
         *reinterpret_cast<short8v_*>(p + base) = pv;
@@ -182,14 +214,18 @@ __global__ void fused_adam_kernel(AdamArgs a) {
         *reinterpret_cast<float4v_*>(m + base + 4) = m1;
         *reinterpret_cast<float4v_*>(v + base) = v0;
         *reinterpret_cast<float4v_*>(v + base + 4) = v1;
+        if (HAS_W) {
+          *reinterpret_cast<float4v_*>(w + base) = w0;
+          *reinterpret_cast<float4v_*>(w + base + 4) = w1;
+        }
       } else {
         for (long j = base; j < n; ++j)
-          adam_elem<T>(p, g, m, v, j, a, inv_bc1, inv_bc2);
+          adam_elem<T, HAS_W>(p, g, m, v, w, j, a, inv_bc1, inv_bc2);
       }
     } else {
       const long end = (base + GRP <= n) ? base + GRP : n;
       for (long j = base; j < end; ++j)
-        adam_elem<T>(p, g, m, v, j, a, inv_bc1, inv_bc2);
+        adam_elem<T, HAS_W>(p, g, m, v, w, j, a, inv_bc1, inv_bc2);
     }
   }
 }
@@ -204,10 +240,12 @@ static long groups_of(long numel) { return (numel + GRP - 1) / GRP; }
 template <typename scalar_t>
 static void sgd_launch(std::vector<at::Tensor>& params,
                        std::vector<at::Tensor>& grads,
-                       std::vector<at::Tensor>& moms, double lr,
+                       std::vector<at::Tensor>& moms,
+                       std::vector<at::Tensor>& masters, double lr,
                        double momentum, double weight_decay) {
   auto stream = at::hip::getCurrentHIPStream();
   const bool has_m = !moms.empty();
+  const bool has_w = !masters.empty();
   for (size_t base = 0; base < params.size(); base += MAX_T) {
     SgdArgs a{};
     a.n_tensors = (int)std::min<size_t>(MAX_T, params.size() - base);
@@ -220,44 +258,58 @@ static void sgd_launch(std::vector<at::Tensor>& params,
       a.p[k] = params[base + k].data_ptr();
       a.g[k] = grads[base + k].data_ptr();
       a.m[k] = has_m ? moms[base + k].data_ptr() : nullptr;
+      a.w[k] = has_w ? masters[base + k].data_ptr() : nullptr;
       a.numel[k] = params[base + k].numel();
       cum += groups_of(a.numel[k]);
       a.cum[k + 1] = cum;
     }
     const int block = 256;
     dim3 grid(grid_for(cum, block));
-    if (has_m)
-      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, true>), grid, dim3(block),
-                         0, stream.stream(), a);
+    if (has_m && has_w)
+      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, true, true>), grid,
+                         dim3(block), 0, stream.stream(), a);
+    else if (has_m)
+      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, true, false>), grid,
+                         dim3(block), 0, stream.stream(), a);
+    else if (has_w)
+      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, false, true>), grid,
+                         dim3(block), 0, stream.stream(), a);
     else
-      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, false>), grid,
+      hipLaunchKernelGGL((fused_sgd_kernel<scalar_t, false, false>), grid,
                          dim3(block), 0, stream.stream(), a);
   }
 }
 
 void fused_sgd(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
-               std::vector<at::Tensor> moms, double lr, double momentum,
-               double weight_decay) {
+               std::vector<at::Tensor> moms, std::vector<at::Tensor> masters,
+               double lr, double momentum, double weight_decay) {
   TORCH_CHECK(!params.empty(), "no params");
   TORCH_CHECK(params.size() == grads.size());
   TORCH_CHECK(moms.empty() || moms.size() == params.size());
+  TORCH_CHECK(masters.empty() || masters.size() == params.size());
   for (size_t i = 0; i < params.size(); ++i) {
     TORCH_CHECK(params[i].is_contiguous() && grads[i].is_contiguous());
     TORCH_CHECK(params[i].scalar_type() == grads[i].scalar_type());
     if (!moms.empty())
       TORCH_CHECK(moms[i].scalar_type() == at::kFloat,
                   "momentum buffers must be fp32");
+    if (!masters.empty())
+      TORCH_CHECK(masters[i].scalar_type() == at::kFloat &&
+                      masters[i].is_contiguous(),
+                  "master weights must be contiguous fp32");
   }
   switch (params[0].scalar_type()) {
     case at::kBFloat16:
-      sgd_launch<c10::BFloat16>(params, grads, moms, lr, momentum,
+      sgd_launch<c10::BFloat16>(params, grads, moms, masters, lr, momentum,
                                 weight_decay);
       break;
     case at::kFloat:
-      sgd_launch<float>(params, grads, moms, lr, momentum, weight_decay);
+      sgd_launch<float>(params, grads, moms, masters, lr, momentum,
+                        weight_decay);
       break;
     case at::kHalf:
-      sgd_launch<c10::Half>(params, grads, moms, lr, momentum, weight_decay);
+      sgd_launch<c10::Half>(params, grads, moms, masters, lr, momentum,
+                            weight_decay);
       break;
     default:
       TORCH_CHECK(false, "fused_sgd: unsupported dtype");
@@ -268,10 +320,12 @@ template <typename scalar_t>
 static void adam_launch(std::vector<at::Tensor>& params,
                         std::vector<at::Tensor>& grads,
                         std::vector<at::Tensor>& ms,
-                        std::vector<at::Tensor>& vs, double lr, double beta1,
-                        double beta2, double eps, double weight_decay,
-                        double bc1, double bc2) {
+                        std::vector<at::Tensor>& vs,
+                        std::vector<at::Tensor>& masters, double lr,
+                        double beta1, double beta2, double eps,
+                        double weight_decay, double bc1, double bc2) {
   auto stream = at::hip::getCurrentHIPStream();
+  const bool has_w = !masters.empty();
   for (size_t base = 0; base < params.size(); base += MAX_T) {
     AdamArgs a{};
     a.n_tensors = (int)std::min<size_t>(MAX_T, params.size() - base);
@@ -289,38 +343,49 @@ static void adam_launch(std::vector<at::Tensor>& params,
       a.g[k] = grads[base + k].data_ptr();
       a.m[k] = ms[base + k].data_ptr();
       a.v[k] = vs[base + k].data_ptr();
+      a.w[k] = has_w ? masters[base + k].data_ptr() : nullptr;
       a.numel[k] = params[base + k].numel();
       cum += groups_of(a.numel[k]);
       a.cum[k + 1] = cum;
     }
     const int block = 256;
     dim3 grid(grid_for(cum, block));
-    hipLaunchKernelGGL((fused_adam_kernel<scalar_t>), grid, dim3(block), 0,
-                       stream.stream(), a);
+    if (has_w)
+      hipLaunchKernelGGL((fused_adam_kernel<scalar_t, true>), grid,
+                         dim3(block), 0, stream.stream(), a);
+    else
+      hipLaunchKernelGGL((fused_adam_kernel<scalar_t, false>), grid,
+                         dim3(block), 0, stream.stream(), a);
   }
 }
 
 void fused_adam(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                 std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                double lr, double beta1, double beta2, double eps,
-                double weight_decay, double bc1, double bc2) {
+                std::vector<at::Tensor> masters, double lr, double beta1,
+                double beta2, double eps, double weight_decay, double bc1,
+                double bc2) {
   TORCH_CHECK(!params.empty(), "no params");
   TORCH_CHECK(params.size() == grads.size() && params.size() == ms.size() &&
               params.size() == vs.size());
+  TORCH_CHECK(masters.empty() || masters.size() == params.size());
   for (size_t i = 0; i < params.size(); ++i) {
     TORCH_CHECK(params[i].is_contiguous() && grads[i].is_contiguous());
     TORCH_CHECK(ms[i].scalar_type() == at::kFloat &&
                 vs[i].scalar_type() == at::kFloat,
                 "Adam moments must be fp32");
+    if (!masters.empty())
+      TORCH_CHECK(masters[i].scalar_type() == at::kFloat &&
+                      masters[i].is_contiguous(),
+                  "master weights must be contiguous fp32");
   }
   switch (params[0].scalar_type()) {
     case at::kBFloat16:
-      adam_launch<c10::BFloat16>(params, grads, ms, vs, lr, beta1, beta2, eps,
-                                 weight_decay, bc1, bc2);
+      adam_launch<c10::BFloat16>(params, grads, ms, vs, masters, lr, beta1,
+                                 beta2, eps, weight_decay, bc1, bc2);
       break;
     case at::kFloat:
-      adam_launch<float>(params, grads, ms, vs, lr, beta1, beta2, eps,
-                         weight_decay, bc1, bc2);
+      adam_launch<float>(params, grads, ms, vs, masters, lr, beta1, beta2,
+                         eps, weight_decay, bc1, bc2);
       break;
     default:
       TORCH_CHECK(false, "fused_adam: unsupported dtype");

@@ -3,9 +3,17 @@
 On GPU these call the hand-written CDNA4 multi-tensor-apply kernels in
 ``saturn_amd._C`` (one launch updates every parameter chunk; HBM-bound, so
 the kernel is a vectorized grid-stride sweep).  On CPU they fall back to
-torch ``_foreach_`` ops so the orchestration test suite runs in the no-GPU
-container.  The reference uses stock ``torch.optim.SGD``
-(simple-verification.py:59); fused Adam is required by the north star.
+fp32-accumulating torch math so the orchestration test suite runs in the
+no-GPU container with the same numerics contract as the kernel:
+
+- momentum buffers and Adam moments are ALWAYS fp32 regardless of the
+  parameter dtype (the HIP kernel requires it);
+- ``master_weights=True`` keeps an fp32 master copy per low-precision
+  parameter: the fused kernel updates the master in fp32 and writes the
+  rounded bf16/fp16 copy in the same launch, avoiding long-horizon update
+  cancellation when ``lr * grad`` drops below bf16 resolution.  The
+  reference trains fp32 models (simple-verification.py:59) and never had
+  this problem.
 """
 
 from __future__ import annotations
@@ -35,9 +43,21 @@ class FusedSGD(torch.optim.Optimizer):
         lr: float,
         momentum: float = 0.0,
         weight_decay: float = 0.0,
+        master_weights: bool = False,
     ) -> None:
-        defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay)
+        defaults = dict(
+            lr=lr,
+            momentum=momentum,
+            weight_decay=weight_decay,
+            master_weights=master_weights,
+        )
         super().__init__(params, defaults)
+
+    def _master_of(self, p: torch.nn.Parameter) -> Optional[torch.Tensor]:
+        st = self.state[p]
+        if "master" not in st:
+            st["master"] = p.data.detach().to(torch.float32).contiguous()
+        return st["master"]
 
     @torch.no_grad()
     def step(self, closure=None) -> Optional[float]:
@@ -46,40 +66,55 @@ class FusedSGD(torch.optim.Optimizer):
             lr = group["lr"]
             mom = group["momentum"]
             wd = group["weight_decay"]
+            use_master = group.get("master_weights", False)
             ps = [p for p in group["params"] if p.grad is not None]
-            for (device, _dtype), chunk in _grouped(ps).items():
+            for (device, dtype), chunk in _grouped(ps).items():
                 grads = [p.grad for p in chunk]
                 if mom != 0.0:
                     bufs = []
                     for p in chunk:
                         st = self.state[p]
                         if "momentum_buffer" not in st:
-                            st["momentum_buffer"] = torch.zeros_like(p)
+                            # fp32 regardless of param dtype: the HIP kernel
+                            # requires fp32 momentum (fused_optim.hip) and
+                            # low-precision momentum accumulation diverges
+                            st["momentum_buffer"] = torch.zeros_like(
+                                p, dtype=torch.float32
+                            )
                         bufs.append(st["momentum_buffer"])
                 else:
                     bufs = None
+                masters = None
+                if use_master and dtype != torch.float32:
+                    masters = [self._master_of(p) for p in chunk]
                 if device.type == "cuda":
                     ext = require_ext()
                     ext.fused_sgd(
                         [p.data for p in chunk],
                         grads,
                         bufs if bufs is not None else [],
+                        masters if masters is not None else [],
                         lr,
                         mom,
                         wd,
                     )
                 else:
-                    if wd != 0.0:
-                        torch._foreach_add_(
-                            grads, [p.data for p in chunk], alpha=wd
+                    # fp32-accumulating fallback, same math as the kernel
+                    for i, p in enumerate(chunk):
+                        g32 = grads[i].float()
+                        pv = (
+                            masters[i]
+                            if masters is not None
+                            else p.data.float()
                         )
-                    if bufs is not None:
-                        torch._foreach_mul_(bufs, mom)
-                        torch._foreach_add_(bufs, grads)
-                        grads = bufs
-                    torch._foreach_add_(
-                        [p.data for p in chunk], grads, alpha=-lr
-                    )
+                        if wd != 0.0:
+                            g32 = g32.add(pv, alpha=wd)
+                        if bufs is not None:
+                            bufs[i].mul_(mom).add_(g32)
+                            g32 = bufs[i]
+                        nv = pv.add_(g32, alpha=-lr) if masters is not None \
+                            else pv.add(g32, alpha=-lr)
+                        p.data.copy_(nv.to(p.dtype))
         return loss
 
 
@@ -93,8 +128,15 @@ class FusedAdam(torch.optim.Optimizer):
         betas=(0.9, 0.999),
         eps: float = 1e-8,
         weight_decay: float = 0.0,
+        master_weights: bool = False,
     ) -> None:
-        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        defaults = dict(
+            lr=lr,
+            betas=betas,
+            eps=eps,
+            weight_decay=weight_decay,
+            master_weights=master_weights,
+        )
         super().__init__(params, defaults)
 
     @torch.no_grad()
@@ -105,9 +147,11 @@ class FusedAdam(torch.optim.Optimizer):
             beta1, beta2 = group["betas"]
             eps = group["eps"]
             wd = group["weight_decay"]
+            use_master = group.get("master_weights", False)
             ps = [p for p in group["params"] if p.grad is not None]
-            for (device, _dtype), chunk in _grouped(ps).items():
+            for (device, dtype), chunk in _grouped(ps).items():
                 m, v, steps = [], [], []
+                masters = [] if use_master and dtype != torch.float32 else None
                 for p in chunk:
                     st = self.state[p]
                     if "step" not in st:
@@ -115,9 +159,13 @@ class FusedAdam(torch.optim.Optimizer):
                         # fp32 moments regardless of param dtype (bf16-safe)
                         st["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
                         st["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                    if masters is not None and "master" not in st:
+                        st["master"] = p.data.detach().to(torch.float32).contiguous()
                     st["step"] += 1
                     m.append(st["exp_avg"])
                     v.append(st["exp_avg_sq"])
+                    if masters is not None:
+                        masters.append(st["master"])
                     steps.append(st["step"])
                 step_t = steps[0]  # uniform within a chunk in practice
                 bc1 = 1.0 - beta1**step_t
@@ -130,6 +178,7 @@ class FusedAdam(torch.optim.Optimizer):
                         grads,
                         m,
                         v,
+                        masters if masters is not None else [],
                         lr,
                         beta1,
                         beta2,
@@ -144,10 +193,12 @@ class FusedAdam(torch.optim.Optimizer):
                     torch._foreach_add_(m, gf, alpha=1 - beta1)
                     torch._foreach_mul_(v, beta2)
                     torch._foreach_addcmul_(v, gf, gf, value=1 - beta2)
-                    for p, mi, vi in zip(chunk, m, v):
+                    for i, (p, mi, vi) in enumerate(zip(chunk, m, v)):
                         denom = (vi / bc2).sqrt_().add_(eps)
                         upd = (mi / bc1).div_(denom)
+                        pv = masters[i] if masters is not None else p.data.float()
                         if wd != 0.0:
-                            p.data.mul_(1 - lr * wd)
-                        p.data.add_(upd.to(p.dtype), alpha=-lr)
+                            pv.mul_(1 - lr * wd)
+                        pv.add_(upd, alpha=-lr)
+                        p.data.copy_(pv.to(p.dtype))
         return loss

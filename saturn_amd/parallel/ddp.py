@@ -40,6 +40,7 @@ class _Bucket:
             offset += n
         self.pending = 0
         self.work: Optional[dist.Work] = None
+        self.launched = False  # native-comm path has no Work handle
 
     def attach_grads(self) -> None:
         for p, v in zip(self.params, self.views):
@@ -48,6 +49,7 @@ class _Bucket:
     def reset(self) -> None:
         self.pending = len(self.params)
         self.work = None
+        self.launched = False
 
 
 class BucketedDDP(torch.nn.Module):
@@ -132,22 +134,42 @@ class BucketedDDP(torch.nn.Module):
             _Bucket(params, p0.device, grad_dtype or p0.dtype)
         )
 
+    def _launch(self, b: _Bucket) -> None:
+        if self.comm is not None:
+            # enqueued on the native engine's comm stream, fenced
+            # against the producing compute stream — overlaps backward
+            self.comm.all_reduce(b.flat, True)
+            b.launched = True
+        else:
+            b.work = dist.all_reduce(b.flat, async_op=True, group=self.pg)
+
     def _hook(self, p: torch.nn.Parameter) -> None:
         b = self._param_bucket[p]
         b.pending -= 1
         if b.pending == 0 and self.world > 1:
-            if self.comm is not None:
-                # enqueued on the native engine's comm stream, fenced
-                # against the producing compute stream — overlaps backward
-                self.comm.all_reduce(b.flat, True)
-            else:
-                b.work = dist.all_reduce(b.flat, async_op=True, group=self.pg)
+            self._launch(b)
 
     def forward(self, *args, **kwargs):
         return self.module(*args, **kwargs)
 
     def grad_sync(self) -> None:
-        """Wait for in-flight bucket all-reduces and average."""
+        """Wait for in-flight bucket all-reduces and average.
+
+        A bucket whose params received no grad this step (``pending > 0``,
+        e.g. an unused head) is still all-reduced here — its flat buffer
+        holds zeros for the missing params, so the average is correct as
+        long as the *set* of grad-less params matches across ranks (it is
+        structural: same model, same step).  Launching here instead of
+        silently scaling an un-reduced bucket fixes the divergence the
+        round-1 advisor flagged.
+        """
+        if self.world > 1:
+            for b in self.buckets:
+                launched = (
+                    b.launched if self.comm is not None else b.work is not None
+                )
+                if not launched:
+                    self._launch(b)
         if self.comm is not None:
             if self.world > 1:
                 self.comm.join()  # compute stream waits the comm stream

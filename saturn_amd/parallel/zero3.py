@@ -119,41 +119,57 @@ class _Unit:
         self._placeholder = torch.empty(0, dtype=dtype, device=device)
         for p in params:
             p.data = self._placeholder
+        # ONE persistent gather buffer whose storage is resized 0 <-> full.
+        # Autograd's saved tensors (weight views captured by the unit's
+        # forward ops) share this storage: resizing to 0 actually frees the
+        # HBM after forward, and the backward re-gather refills the SAME
+        # storage so those saved views become valid again (the trick torch
+        # FSDP uses; merely dropping the Python reference would leave every
+        # unit's full buffer pinned until backward — round-1 advisor
+        # finding).
         self.full: Optional[torch.Tensor] = None
+        self.gathered = False  # gather started (in flight or complete)
         self.work = None
         self.copy_event: Optional["torch.cuda.Event"] = None
         self.grad_pending = 0
 
+    def _materialize_full(self) -> torch.Tensor:
+        if self.full is None:
+            self.full = torch.empty(
+                self.pad_numel, dtype=self.dtype, device=self.device
+            )
+        elif self.full.untyped_storage().size() == 0:
+            self.full.untyped_storage().resize_(
+                self.pad_numel * self.full.element_size()
+            )
+        return self.full
+
     # -- gather / free -----------------------------------------------------
     def start_gather(self, async_op: bool = False) -> None:
-        if self.full is not None:
+        if self.gathered:
             return
+        self.gathered = True
+        full = self._materialize_full()
         cs = _copy_stream(self.device) if self.offload else None
         if cs is not None:
             # pinned-host -> HBM on the copy stream: a prefetched unit's
             # upload overlaps the current unit's compute
-            self.full = torch.empty(
-                self.pad_numel, dtype=self.dtype, device=self.device
-            )
-            self.full.record_stream(cs)  # allocator: buffer is used on cs
+            full.record_stream(cs)  # allocator: buffer is used on cs
             ev = torch.cuda.Event()
             with torch.cuda.stream(cs):
                 shard_dev = self.shard.data.to(self.device, non_blocking=True)
                 if _world() == 1:
-                    self.full.copy_(shard_dev)
+                    full.copy_(shard_dev)
                     self.work = None
                 else:
                     self.work = _allgather_into(
-                        self.full, shard_dev, async_op=async_op
+                        full, shard_dev, async_op=async_op
                     )
                 ev.record(cs)
             self.copy_event = ev
             return
         shard_dev = self.shard.data
-        self.full = torch.empty(
-            self.pad_numel, dtype=self.dtype, device=self.device
-        )
-        self.work = _allgather_into(self.full, shard_dev, async_op=async_op)
+        self.work = _allgather_into(full, shard_dev, async_op=async_op)
 
     def finish_gather(self) -> None:
         if self.copy_event is not None:
@@ -172,7 +188,11 @@ class _Unit:
             # never drop a buffer with an in-flight upload
             self.copy_event.synchronize()
             self.copy_event = None
-        self.full = None
+        if self.work is not None:
+            self.work.wait()
+        if self.full is not None and self.gathered:
+            self.full.untyped_storage().resize_(0)
+        self.gathered = False
         self.work = None
 
     # -- gradient reduce-scatter -------------------------------------------
@@ -347,7 +367,7 @@ class Zero3Model(nn.Module):
     def grad_sync(self) -> None:
         """Finish the step: reduce the residual unit's grads."""
         res = self.units[-1]
-        if res.module is self.model and res.full is not None:
+        if res.module is self.model and res.gathered:
             res.reduce_grads()
             res.free()
 
