@@ -81,23 +81,29 @@ def _ddp_worker(
             except Exception:
                 pass  # optimizer class may have changed between intervals
 
-        def sharded(base_iter):
-            """Round-robin batch sharding across ranks (capability parity
-            with the reference's DistributedSampler dataloader,
-            DDP.py:117-144; resume position under changed world size is
-            approximate there too — SURVEY quirk #11)."""
-            for i, b in enumerate(base_iter):
-                if i % world == rank:
-                    yield b
-
-        it = sharded(
-            task.get_iterator() if not trial else task.get_fresh_iterator()
-        )
+        it = task.get_iterator() if not trial else task.get_fresh_iterator()
 
         last_loss = [0.0]
 
         def step(batch) -> None:
             x, y = batch
+            # Shard the GLOBAL batch across ranks (the reference's
+            # DistributedSampler semantics, DDP.py:117-144): one step
+            # consumes exactly one dataloader batch regardless of world
+            # size, so the parent's cursor advance (reconfigure) and the
+            # solver's runtime model (per-step time drops with more GPUs)
+            # both stay exact.  Round-robin whole batches — the round-1
+            # design — consumed world× batches per step, which the
+            # accounting missed (advisor finding #5).  A batch smaller
+            # than the world is replicated (identical grads; average is a
+            # no-op); a non-divisible batch gives ranks shards differing
+            # by one row, biasing the grad average by O(1/n) — same as
+            # torch's DistributedSampler without padding.
+            n = x.shape[0]
+            if world > 1 and n >= world:
+                lo = rank * n // world
+                hi = (rank + 1) * n // world
+                x, y = x[lo:hi], y[lo:hi]
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
             if x.is_floating_point():
@@ -115,7 +121,7 @@ def _ddp_worker(
             try:
                 return next(it)
             except StopIteration:
-                it = sharded(task.get_fresh_iterator())
+                it = task.get_fresh_iterator()
                 return next(it)
 
         result = None

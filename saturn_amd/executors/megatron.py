@@ -28,7 +28,11 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
     import torch.distributed as dist
 
     from saturn_amd.executors.ddp import _make_optimizer
-    from saturn_amd.parallel.tensor import tp_full_state_dict, tp_shard_model
+    from saturn_amd.parallel.tensor import (
+        tp_full_state_dict,
+        tp_resync_replicated,
+        tp_shard_model,
+    )
 
     backend = init_process_group(rank, world)
     try:
@@ -93,10 +97,14 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             if device.type == "cuda":
                 torch.cuda.synchronize()
             return (timer() - t0) / (TRIAL_BATCHES - 1)
-        for _ in range(batch_count):
+        RESYNC_EVERY = 64  # bound replicated-param drift from fp32 atomics
+        for i in range(batch_count):
             step(next_batch())
+            if (i + 1) % RESYNC_EVERY == 0:
+                tp_resync_replicated(model)
         if device.type == "cuda":
             torch.cuda.synchronize()
+        tp_resync_replicated(model)  # checkpoint from a rank-consistent state
         sd = tp_full_state_dict(model)
         if rank == 0 and sd is not None:
             task.save_checkpoint(sd, None)

@@ -139,6 +139,37 @@ def tp_shard_model(model: nn.Module) -> nn.Module:
     return model
 
 
+def tp_replicated_parameters(model: nn.Module):
+    """Parameters that are replicated (not sharded) under TP: embeddings,
+    norms, lm_head, and row-parallel biases (added post-reduce on every
+    rank).  Column-parallel weight+bias and row-parallel weights are the
+    sharded ones."""
+    sharded = set()
+    for mod in model.modules():
+        if isinstance(mod, ColumnParallelLinear):
+            sharded.add(id(mod.weight))
+            if mod.bias is not None:
+                sharded.add(id(mod.bias))
+        elif isinstance(mod, RowParallelLinear):
+            sharded.add(id(mod.weight))
+    return [p for p in model.parameters() if id(p) not in sharded]
+
+
+@torch.no_grad()
+def tp_resync_replicated(model: nn.Module) -> None:
+    """Broadcast replicated params from rank 0.
+
+    Mathematically every rank computes identical gradients for these, but
+    the fused kernels accumulate with fp32 atomics (embedding scatter-add,
+    attention dQ/dK/dV) in non-deterministic order, so replicated weights
+    drift over many steps (round-1 advisor finding #4).  Call periodically
+    — Megatron broadcasts its non-sharded params the same way."""
+    if _world() == 1:
+        return
+    for p in tp_replicated_parameters(model):
+        dist.broadcast(p.data, src=0)
+
+
 @torch.no_grad()
 def tp_full_state_dict(model: nn.Module):
     """Gather sharded weights back to a full state dict on rank 0."""
