@@ -44,6 +44,12 @@ def main() -> None:
         action="store_true",
         help="capture the step in a hipGraph and replay it (world 1 only)",
     )
+    ap.add_argument(
+        "--tunableop",
+        action="store_true",
+        help="use the offline-tuned hipBLASLt/rocBLAS GEMM algo table "
+        "(tools/tunableop_gfx950.csv), tuning disabled at runtime",
+    )
     args = ap.parse_args()
 
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
@@ -73,6 +79,22 @@ def main() -> None:
         from saturn_amd.ops import require_ext
 
         require_ext()
+        if args.tunableop:
+            # READ-ONLY use of the offline-tuned GEMM algo table; runtime
+            # tuning stays off so the timed region is never polluted
+            csv = os.path.join(
+                os.path.dirname(os.path.abspath(__file__)),
+                "tools", "tunableop_gfx950.csv",
+            )
+            if os.path.isfile(csv):
+                import torch.cuda.tunable as tunable
+
+                tunable.enable(True)
+                tunable.tuning_enable(False)
+                tunable.read_file(csv)
+                log(f"[rank {rank}] tunableop: {len(tunable.get_results())} tuned GEMMs loaded")
+            else:
+                log(f"[rank {rank}] tunableop csv missing at {csv}; skipping")
 
     from saturn_amd.ops.optim import FusedSGD
     from saturn_amd.parallel.ddp import BucketedDDP

@@ -65,6 +65,7 @@ class BucketedDDP(torch.nn.Module):
         grad_dtype: Optional[torch.dtype] = None,
         comm=None,
         exclude=None,
+        use_buckets: Optional[bool] = None,
     ) -> None:
         """``comm``: an optional saturn_amd native RcclComm; when given,
         bucket all-reduces run on its dedicated HIP stream (ncclAvg)
@@ -103,9 +104,19 @@ class BucketedDDP(torch.nn.Module):
             p for p in module.parameters()
             if p.requires_grad and id(p) not in skip
         ]
-        bucket_bytes = int(bucket_mb * 1024 * 1024)
         self.buckets: List[_Bucket] = []
         self._param_bucket = {}
+        if use_buckets is None:
+            use_buckets = self.world > 1
+        if not use_buckets:
+            # No comm -> no buckets.  Grad-as-bucket-view costs ~4 bytes/um
+            # of pure bookkeeping traffic per step (zero the flat + autograd
+            # accumulate-into-view instead of assign); at world 1 the flat
+            # buffer serves nothing, so let autograd assign fresh grads and
+            # zero with set_to_none.  (Measured: ~8 ms/step of FillFunctor +
+            # CUDAFunctor_add on GPT-J-6B, profiles/r02_baseline_kernels.txt)
+            return
+        bucket_bytes = int(bucket_mb * 1024 * 1024)
         cur: List[torch.nn.Parameter] = []
         cur_bytes = 0
         for p in reversed(params):  # backward finishes roughly in reverse
@@ -185,7 +196,11 @@ class BucketedDDP(torch.nn.Module):
             b.reset()
 
     def zero_grad_buffers(self, set_to_none: bool = False) -> None:
-        """Zero the flat grad buffers (grads are views; never set to None)."""
+        """Zero the flat grad buffers (grads are views; never set to None).
+        At world 1 there are no buckets: plain set_to_none zeroing."""
+        if not self.buckets:
+            self.module.zero_grad(set_to_none=True)
+            return
         for b in self.buckets:
             b.flat.zero_()
             b.attach_grads()

@@ -180,7 +180,19 @@ def _greedy_plan(task_list, n_gpus: int) -> Plan:
 # ---------------------------------------------------------------------------
 # MILP formulation
 # ---------------------------------------------------------------------------
-def _milp_plan(task_list, n_gpus: int, timeout: float) -> Optional[Plan]:
+def _milp_plan(
+    task_list,
+    n_gpus: int,
+    timeout: float,
+    incumbent_bound: Optional[float] = None,
+) -> Optional[Plan]:
+    """``incumbent_bound``: an upper bound on the makespan of any plan worth
+    adopting (previous plan's remaining makespan minus the hysteresis
+    margin).  scipy's HiGHS interface cannot take a warm-start solution
+    (the reference warm-starts Gurobi, milp.py:322-327), so the bound is
+    how the previous incumbent prunes the branch-and-bound tree; when no
+    plan beats it the model is infeasible, which HiGHS usually proves
+    quickly, and the caller keeps the shifted previous plan."""
     T = len(task_list)
     if T == 0:
         return Plan([], [], [], [], [], [], 0.0, "empty")
@@ -224,6 +236,8 @@ def _milp_plan(task_list, n_gpus: int, timeout: float) -> Optional[Plan]:
         ub[v] = 1
     ub[start_off : start_off + T] = horizon
     ub[mk] = horizon
+    if incumbent_bound is not None:
+        ub[mk] = min(horizon, float(incumbent_bound))
 
     rows: List[Dict[int, float]] = []
     lo: List[float] = []
@@ -368,15 +382,40 @@ def solve(
     plan is adopted only if its makespan beats the saved plan's remaining
     makespan by more than ``interval + hysteresis`` (reference
     milp.py:363-381); otherwise the saved plan is shifted by ``interval``.
+
+    Warm start: the previous plan's remaining makespan (minus the
+    hysteresis margin) is fed to the MILP as an upper bound on the
+    makespan variable — the scipy/HiGHS equivalent of the reference's
+    Gurobi ``warmStart`` incumbent (milp.py:322-327): the tree is pruned
+    to only plans worth adopting, and on timeout/infeasibility the
+    *shifted previous plan* is kept instead of degrading to greedy.
     """
     if n_gpus is None:
         n_gpus = detect_gpu_count()
-    plan = _milp_plan(task_list, n_gpus, timeout)
+    names = [t.name for t in task_list]
+    can_keep = presolved is not None and presolved.task_names == names
+    covers = presolved is not None and set(presolved.task_names) >= set(names)
+    bound = None
+    if covers:
+        bound = max(0.0, presolved.makespan - interval - hysteresis)
+    plan = _milp_plan(task_list, n_gpus, timeout, incumbent_bound=bound)
     if plan is None:
-        log.warning("MILP produced no incumbent within %.0fs; greedy fallback", timeout)
-        plan = _greedy_plan(task_list, n_gpus)
+        if covers:
+            kept = presolved if can_keep else presolved.restrict(names)
+            kept.shift(interval)
+            kept.solver_status = "kept_incumbent"
+            log.info(
+                "no plan beats the previous one within %.0fs; keeping the "
+                "shifted incumbent (makespan %.1f)", timeout, kept.makespan,
+            )
+            return kept
+        log.warning(
+            "MILP produced no incumbent within %.0fs; greedy fallback",
+            timeout,
+        )
+        return _greedy_plan(task_list, n_gpus)
 
-    if presolved is not None and presolved.task_names == plan.task_names:
+    if can_keep:
         if plan.makespan < presolved.makespan - interval - hysteresis:
             return plan
         kept = presolved

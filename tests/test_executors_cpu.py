@@ -132,7 +132,9 @@ def test_ddp_buckets_partition_params_exactly(sizes, bucket_mb):
                 setattr(self, f"p{i}", torch.nn.Parameter(torch.randn(n)))
 
     m = M()
-    ddp = BucketedDDP(m, bucket_mb=bucket_mb)
+    # world 1 skips buckets by default (no comm to feed); force them here
+    # to exercise the partitioning logic the world>1 path uses
+    ddp = BucketedDDP(m, bucket_mb=bucket_mb, use_buckets=True)
     seen = set()
     for b in ddp.buckets:
         for p in b.params:

@@ -61,6 +61,68 @@ def test_fused_sgd_bf16():
     assert rel_err(p, ref.to(torch.bfloat16)) < 1e-2
 
 
+def test_fused_sgd_bf16_momentum():
+    """Round-1 latent bug: bf16 params + momentum crashed (momentum buffer
+    was allocated in param dtype; the kernel requires fp32)."""
+    requires_ext()
+    from saturn_amd.ops.optim import FusedSGD
+
+    torch.manual_seed(0)
+    p32 = torch.randn(4096, device="cuda")
+    p = p32.to(torch.bfloat16).requires_grad_(True)
+    ref = p32.clone().requires_grad_(True)
+    opt = FusedSGD([p], lr=0.05, momentum=0.9)
+    ref_opt = torch.optim.SGD([ref], lr=0.05, momentum=0.9)
+    for _ in range(4):
+        g = torch.randn(4096, device="cuda")
+        p.grad = g.to(torch.bfloat16)
+        ref.grad = g.to(torch.bfloat16).float()
+        opt.step()
+        ref_opt.step()
+    assert opt.state[p]["momentum_buffer"].dtype == torch.float32
+    assert rel_err(p, ref.to(torch.bfloat16)) < 3e-2
+
+
+def test_fused_sgd_master_weights_gpu():
+    """The fused kernel's fp32 master path: sub-bf16-resolution updates
+    must accumulate in the master and round into the bf16 param."""
+    requires_ext()
+    from saturn_amd.ops.optim import FusedSGD
+
+    steps, lr = 40, 1e-4
+    p = torch.ones(4096, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    opt = FusedSGD([p], lr=lr, master_weights=True)
+    for _ in range(steps):
+        p.grad = torch.ones_like(p)
+        opt.step()
+    mw = opt.state[p]["master"]
+    expect = 1.0 - steps * lr
+    assert torch.allclose(mw, torch.full_like(mw, expect), atol=1e-5)
+    assert torch.equal(p.data, mw.to(torch.bfloat16))
+
+
+def test_fused_adam_master_weights_gpu():
+    requires_ext()
+    from saturn_amd.ops.optim import FusedAdam
+
+    torch.manual_seed(0)
+    init = torch.randn(2048, device="cuda")
+    p32 = init.clone().requires_grad_(True)
+    ref = FusedAdam([p32], lr=1e-3, weight_decay=0.01)
+    pb = init.to(torch.bfloat16).requires_grad_(True)
+    ours = FusedAdam([pb], lr=1e-3, weight_decay=0.01, master_weights=True)
+    for _ in range(5):
+        g = torch.randn(2048, device="cuda")
+        p32.grad = g.clone()
+        ref.step()
+        pb.grad = g.to(torch.bfloat16)
+        ours.step()
+    mw = ours.state[pb]["master"]
+    # differs from the fp32 run only by bf16 gradient rounding
+    assert rel_err(mw, p32) < 3e-2
+    assert torch.equal(pb.data, mw.to(torch.bfloat16))
+
+
 def test_fused_adam_matches_adamw():
     requires_ext()
     from saturn_amd.ops.optim import FusedAdam
