@@ -24,6 +24,7 @@ import torch
 import torch.nn as nn
 
 from saturn_amd.ops.functional import (
+    fused_gelu,
     FusedLayerNorm,
     apply_rope,
     causal_attention,
@@ -58,9 +59,16 @@ class GPTJAttention(nn.Module):
         self.k_proj = nn.Linear(E, E, bias=False)
         self.v_proj = nn.Linear(E, E, bias=False)
         self.out_proj = nn.Linear(E, E, bias=False)
-        cos, sin = rope_tables(cfg.n_ctx, cfg.rotary_dim)
-        self.register_buffer("rope_cos", cos, persistent=False)
-        self.register_buffer("rope_sin", sin, persistent=False)
+        self.n_ctx = cfg.n_ctx
+        # NOT buffers: model.to(bf16) would quantize the angle tables, and
+        # the fused kernel wants the same fp32 [T, half] rows every call
+        # (no per-call dtype convert / expand); device placement is lazy.
+        self._rope_f32 = rope_tables(cfg.n_ctx, cfg.rotary_dim)
+
+    def _rope(self, device):
+        if self._rope_f32[0].device != device:
+            self._rope_f32 = tuple(t.to(device) for t in self._rope_f32)
+        return self._rope_f32
 
     def forward(self, x):
         B, T, E = x.shape
@@ -68,9 +76,9 @@ class GPTJAttention(nn.Module):
         q = self.q_proj(x).view(B, T, H, D)
         k = self.k_proj(x).view(B, T, H, D)
         v = self.v_proj(x).view(B, T, H, D)
-        cos, sin = self.rope_cos, self.rope_sin  # apply_rope slices positions
-        # partial rotary: the kernel rotates the first rotary_dim dims in
-        # place and passes the rest through — no split+cat (the reference
+        cos, sin = self._rope(x.device)  # fp32 [n_ctx, half]
+        # partial rotary: the kernel rotates the first rotary_dim dims and
+        # copies the rest through — no split+cat (the reference
         # materializes both halves and concatenates, GPTJ.py:255-259)
         q = apply_rope(q, cos, sin)
         k = apply_rope(k, cos, sin)
@@ -90,7 +98,7 @@ class GPTJMLP(nn.Module):
         self.fc_out = nn.Linear(4 * E, E)
 
     def forward(self, x):
-        return self.fc_out(torch.nn.functional.gelu(self.fc_in(x), approximate="tanh"))
+        return self.fc_out(fused_gelu(self.fc_in(x)))
 
 
 class GPTJBlock(nn.Module):

@@ -57,9 +57,15 @@ class LlamaAttention(nn.Module):
         self.k_proj = nn.Linear(E, self.n_kv * self.head_dim, bias=False)
         self.v_proj = nn.Linear(E, self.n_kv * self.head_dim, bias=False)
         self.o_proj = nn.Linear(self.n_head * self.head_dim, E, bias=False)
-        cos, sin = rope_tables(cfg.n_ctx, self.head_dim, base=cfg.rope_theta)
-        self.register_buffer("rope_cos", cos, persistent=False)
-        self.register_buffer("rope_sin", sin, persistent=False)
+        # NOT buffers: model.to(bf16) would quantize the angle tables (see
+        # gptj.py); fp32 [T, half] rows read directly by the fused kernel
+        self._rope_f32 = rope_tables(cfg.n_ctx, self.head_dim,
+                                     base=cfg.rope_theta)
+
+    def _rope(self, device):
+        if self._rope_f32[0].device != device:
+            self._rope_f32 = tuple(t.to(device) for t in self._rope_f32)
+        return self._rope_f32
 
     def forward(self, x):
         B, T, E = x.shape
@@ -67,7 +73,7 @@ class LlamaAttention(nn.Module):
         q = self.q_proj(x).view(B, T, self.n_head, D)
         k = self.k_proj(x).view(B, T, self.n_kv, D)
         v = self.v_proj(x).view(B, T, self.n_kv, D)
-        cos, sin = self.rope_cos, self.rope_sin  # apply_rope slices positions
+        cos, sin = self._rope(x.device)
         q = apply_rope(q, cos, sin, half_style=True)
         k = apply_rope(k, cos, sin, half_style=True)
         o = causal_attention(

@@ -22,8 +22,10 @@ std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets,
                                int64_t Tr, int64_t ignore_index);
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                   at::Tensor dloss, int64_t Tr, int64_t ignore_index);
-void rope_apply(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
-                bool half_style, bool backward);
+void rope_apply(at::Tensor y, at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
+                int64_t t_len, int64_t t_off, bool half_style, bool backward);
+at::Tensor gelu_fwd(at::Tensor x);
+at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                  bool causal);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
@@ -47,7 +49,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("norm_bwd", &samd::norm_bwd, "LayerNorm/RMSNorm backward");
   m.def("ce_fwd", &samd::ce_fwd, "fused cross-entropy forward");
   m.def("ce_bwd", &samd::ce_bwd, "fused cross-entropy backward");
-  m.def("rope_apply", &samd::rope_apply, "fused rotary embedding (in-place)");
+  m.def("rope_apply", &samd::rope_apply,
+        "fused rotary embedding (out-of-place, offset table rows)");
+  m.def("gelu_fwd", &samd::gelu_fwd, "fused tanh-approx GELU forward");
+  m.def("gelu_bwd", &samd::gelu_bwd, "fused tanh-approx GELU backward");
   m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
   m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward");
   m.def("add3", &samd::add3, "fused 3-way residual add");

@@ -161,21 +161,24 @@ def rope_tables(
 
 class _RopeFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, cos_bt, sin_bt, half_style):
+    def forward(ctx, x, cos_t, sin_t, t_len, t_off, half_style):
         ext = require_ext()
-        y = x.contiguous().clone()
-        ext.rope_apply(y, cos_bt, sin_bt, half_style, False)
-        ctx.save_for_backward(cos_bt, sin_bt)
-        ctx.half_style = half_style
+        x = x.contiguous()
+        y = torch.empty_like(x)
+        ext.rope_apply(y, x, cos_t, sin_t, t_len, t_off, half_style, False)
+        ctx.save_for_backward(cos_t, sin_t)
+        ctx.meta = (t_len, t_off, half_style)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = require_ext()
-        cos_bt, sin_bt = ctx.saved_tensors
-        dx = dy.contiguous().clone()
-        ext.rope_apply(dx, cos_bt, sin_bt, ctx.half_style, True)
-        return dx, None, None, None
+        cos_t, sin_t = ctx.saved_tensors
+        t_len, t_off, half_style = ctx.meta
+        dy = dy.contiguous()
+        dx = torch.empty_like(dy)
+        ext.rope_apply(dx, dy, cos_t, sin_t, t_len, t_off, half_style, True)
+        return dx, None, None, None, None, None
 
 
 def _rope_cpu(x, cos_bt, sin_bt, half_style):
@@ -204,10 +207,13 @@ def apply_rope(
     sin: torch.Tensor,
     half_style: bool = False,
 ) -> torch.Tensor:
-    """x: [B, T, H, D]; cos/sin: [T, half] (broadcast over batch).
+    """x: [B, T, H, D]; cos/sin: fp32 [T_total, half] (broadcast over batch
+    and heads — the kernel indexes rows itself, no per-call expand).
 
     GPT-J uses interleaved pairs (half_style=False, GPTJ.py:56-61);
-    Llama/NeoX uses the half-split layout (half_style=True).
+    Llama/NeoX uses the half-split layout (half_style=True).  Under
+    sequence parallelism this rank's shard covers global positions
+    [rank*T, rank*T + T) — the kernel reads table rows at that offset.
     """
     B, T, H, D = x.shape
     half = cos.shape[-1]
@@ -219,14 +225,16 @@ def apply_rope(
             "rope table too short for this sequence shard — pass the FULL "
             "table to apply_rope (it slices positions itself)"
         )
-    cos = cos[off : off + T]
-    sin = sin[off : off + T]
-    # model.to(dtype=bf16) converts registered buffers — force fp32 tables
-    cos_bt = cos.float().unsqueeze(0).expand(B, T, half).contiguous()
-    sin_bt = sin.float().unsqueeze(0).expand(B, T, half).contiguous()
+    if cos.dtype != torch.float32:
+        # tables must stay fp32 (model.to(bf16) would quantize angles)
+        cos = cos.float()
+        sin = sin.float()
     if x.is_cuda:
-        return _RopeFn.apply(x, cos_bt, sin_bt, half_style)
-    return _rope_cpu(x, cos_bt, sin_bt, half_style)
+        return _RopeFn.apply(x, cos.contiguous(), sin.contiguous(), T, off,
+                             half_style)
+    cos_bt = cos[off : off + T].unsqueeze(0).expand(B, T, half)
+    sin_bt = sin[off : off + T].unsqueeze(0).expand(B, T, half)
+    return _rope_cpu(x, cos_bt.contiguous(), sin_bt.contiguous(), half_style)
 
 
 # ---------------------------------------------------------------------------
@@ -298,6 +306,29 @@ def fused_add3(a, b, c):
     if a.is_cuda:
         return _Add3Fn.apply(a, b, c)
     return a + b + c
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ext = require_ext()
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return ext.gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = require_ext()
+        (x,) = ctx.saved_tensors
+        return ext.gelu_bwd(dy.contiguous(), x)
+
+
+def fused_gelu(x: torch.Tensor) -> torch.Tensor:
+    """tanh-approx GELU (GPT-J's activation, GPTJ.py:25-41) as one
+    vectorized CDNA4 kernel each way; torch fallback on CPU / fp32."""
+    if x.is_cuda and x.dtype in (torch.bfloat16, torch.float16):
+        return _GeluFn.apply(x)
+    return torch.nn.functional.gelu(x, approximate="tanh")
 
 
 class _SwigluFn(torch.autograd.Function):

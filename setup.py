@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(CSRC, "layernorm.hip"),
         os.path.join(CSRC, "cross_entropy.hip"),
         os.path.join(CSRC, "rope.hip"),
+        os.path.join(CSRC, "gelu.hip"),
         os.path.join(CSRC, "attention.hip"),
         os.path.join(CSRC, "add3.hip"),
         os.path.join(CSRC, "swiglu.hip"),
