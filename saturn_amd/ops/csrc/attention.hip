@@ -686,13 +686,12 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 const unsigned short* __restrict__ V,
                                 const float* __restrict__ LSE,
                                 const float* __restrict__ DELTA,
-                                float* __restrict__ dQ,
                                 unsigned short* __restrict__ dK,
                                 unsigned short* __restrict__ dV,
                                 float* __restrict__ dKf,
                                 float* __restrict__ dVf, int T, int n_heads,
                                 int n_kv, float scale, int causal, TStr dos,
-                                TStr qs, TStr ks, TStr vs, TStr dqs,
+                                TStr qs, TStr ks, TStr vs,
                                 TStr dks, TStr dvs) {
   using namespace bwd;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -830,43 +829,9 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
       }
     }
 
-    // ---- dQ partials: wave w handles q-subtile (w&1), d-half (w>>1).
-    // NB: names must not shadow the head index h used in the atomic below.
+    // (dQ moved to attn_dq_kernel — q-block grid, register accumulation,
+    // no atomics; this kernel now only accumulates dK/dV.)
     __syncthreads();
-    {
-      const int qsub = wid & 1;
-      const int dhalf = wid >> 1;
-#pragma unroll
-      for (int dt2 = 0; dt2 < D / 32; ++dt2) {
-        const int dt = dhalf * (D / 32) + dt2;
-        f32x4 acc_dq = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-        for (int kt2 = 0; kt2 < 2; ++kt2) {
-          bf16x8 adst, bk;
-          if constexpr (TR16) {
-            tr16_bfrag2<bwd::QT + 8, D + 8>(
-                &lds.dst[kt2 * 32 + qg * 8][qsub * 16],
-                &lds.kt[kt2 * 32 + qg * 8][dt * 16], lane, adst, bk);
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j) {
-              adst[j] = (short)lds.dst[kt2 * 32 + qg * 8 + j][qsub * 16 + r];
-              bk[j] = (short)lds.kt[kt2 * 32 + qg * 8 + j][dt * 16 + r];
-            }
-          }
-          // A[q row][key k]: dst is [key][q]; the transposed read above
-          // gives lane l -> A[l&15 q][(l>>4)*8+j key] as required.
-          acc_dq = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bk, acc_dq, 0, 0, 0);
-        }
-#pragma unroll
-        for (int reg = 0; reg < 4; ++reg) {
-          const int q_glob = q0 + qsub * 16 + qg * 4 + reg;
-          atomicAdd(&dQ[b * dqs.sb + h * dqs.sh + (long)q_glob * dqs.st +
-                        dt * 16 + r],
-                    acc_dq[reg]);
-        }
-      }
-    }
   }
 
   // ---- epilogue: write dK, dV.  With GQA (several q heads per kv head)
@@ -901,6 +866,205 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
   }
 }
 
+// ---------------------------------------------------------------------------
+// dQ kernel (backward v2, round 2).  The combined kernel's dQ phase was
+// 41-54% of backward time, and its fp32 atomics alone 23-37% (measured via
+// the dqmode probe; every dQ element was atomically added by T/KB
+// kv-blocks).  This kernel re-parallelizes dQ over Q-BLOCKS: each block
+// owns 64 q rows, loops over the key tiles, recomputes S^T and dP^T
+// (+2 GEMM units of recompute vs the combined kernel's shared tiles), and
+// accumulates dQ in registers — no atomics, no fp32 scratch buffer, no
+// bf16 cast pass; the result is stored once as bf16 in the [B,T,H,D]
+// physical layout the upstream projections use.
+//
+// Routing: S^T / dS^T land in the MFMA D-layout with key on the reg dim
+// and q on l&15.  The dQ MFMA wants A[m=q][k=key over 32] — exactly the
+// cross-quarter exchange validated in tools/bwd_route_probe.hip: pack the
+// 4 f32 regs as 2 u32 of bf16 pairs, one __shfl at lane stride 16 per
+// pair, assemble 8 slots.  No LDS round-trip for dS^T.
+// ---------------------------------------------------------------------------
+namespace dq {
+constexpr int QB = 64;  // q rows per block (4 waves x 16)
+constexpr int KT = 32;  // keys per tile iteration
+}  // namespace dq
+
+template <int D>
+struct AttnDqLds {
+  unsigned short qt[dq::QB][D + 8];
+  unsigned short dot[dq::QB][D + 8];
+  unsigned short kt[dq::KT][D + 8];
+  unsigned short vt[dq::KT][D + 8];
+  float lse_t[dq::QB];
+  float delta_t[dq::QB];
+};
+
+template <int D, bool TR16>
+__launch_bounds__(256, D <= 128 ? 2 : 1)
+__global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
+                               const unsigned short* __restrict__ Q,
+                               const unsigned short* __restrict__ K,
+                               const unsigned short* __restrict__ V,
+                               const float* __restrict__ LSE,
+                               const float* __restrict__ DELTA,
+                               unsigned short* __restrict__ dQ16, int T,
+                               int n_heads, int n_kv, float scale, int causal,
+                               TStr dos, TStr qs, TStr ks, TStr vs,
+                               TStr dqs) {
+  using namespace dq;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  AttnDqLds<D>& lds = *reinterpret_cast<AttnDqLds<D>*>(smem);
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int r = lane & 15;
+  const int qg = lane >> 4;
+  const int q0 = blockIdx.y * QB;
+  const long bh = blockIdx.x;
+  const long b = bh / n_heads, h = bh % n_heads;
+  const long h_kv = h * n_kv / n_heads;
+  const unsigned short* Qh = Q + b * qs.sb + h * qs.sh;
+  const unsigned short* Kh = K + b * ks.sb + h_kv * ks.sh;
+  const unsigned short* Vh = V + b * vs.sb + h_kv * vs.sh;
+  const unsigned short* dOh = dO + b * dos.sb + h * dos.sh;
+  const float* lse_h = LSE + bh * (long)T;
+  const float* del_h = DELTA + bh * (long)T;
+
+  // ---- stage the block's Q / dO rows + lse/delta once
+  {
+    constexpr int CHUNKS = (QB * D) / (256 * 8);
+#pragma unroll
+    for (int c = 0; c < CHUNKS; ++c) {
+      int idx = (c * 256 + threadIdx.x) * 8;
+      int row = idx / D, col = idx % D;
+      *reinterpret_cast<bf16x8*>(&lds.qt[row][col]) =
+          *reinterpret_cast<const bf16x8*>(Qh + (long)(q0 + row) * qs.st + col);
+      *reinterpret_cast<bf16x8*>(&lds.dot[row][col]) =
+          *reinterpret_cast<const bf16x8*>(
+              dOh + (long)(q0 + row) * dos.st + col);
+    }
+    if (threadIdx.x < QB) {
+      lds.lse_t[threadIdx.x] = lse_h[q0 + threadIdx.x];
+      lds.delta_t[threadIdx.x] = del_h[q0 + threadIdx.x];
+    }
+  }
+
+  f32x4 acc[D / 16];
+#pragma unroll
+  for (int dt = 0; dt < D / 16; ++dt) acc[dt] = {0.f, 0.f, 0.f, 0.f};
+
+  const int k_end = causal ? (q0 + QB < T ? q0 + QB : T) : T;
+  for (int k0 = 0; k0 < k_end; k0 += KT) {
+    __syncthreads();  // previous tile fully consumed (also orders staging)
+    {
+      constexpr int CHUNKS = (KT * D) / (256 * 8);
+#pragma unroll
+      for (int c = 0; c < CHUNKS; ++c) {
+        int idx = (c * 256 + threadIdx.x) * 8;
+        int row = idx / D, col = idx % D;
+        *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
+            *reinterpret_cast<const bf16x8*>(
+                Kh + (long)(k0 + row) * ks.st + col);
+        *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
+            *reinterpret_cast<const bf16x8*>(
+                Vh + (long)(k0 + row) * vs.st + col);
+      }
+    }
+    __syncthreads();
+
+    // Waves whose 16 q rows are entirely left of this key tile contribute
+    // nothing (causal) — skip the math, keep the barriers block-uniform.
+    if (causal && k0 > q0 + wid * 16 + 15) continue;
+
+    // ---- per 16-key subtile: S^T, dP^T -> dS^T (key on reg, q on lane)
+    float ds_sub[2][4];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      f32x4 st = {0.f, 0.f, 0.f, 0.f};
+      f32x4 dpt = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ds_ = 0; ds_ < D / 32; ++ds_) {
+        bf16x8 ak = *reinterpret_cast<const bf16x8*>(
+            &lds.kt[n * 16 + r][ds_ * 32 + qg * 8]);
+        bf16x8 av = *reinterpret_cast<const bf16x8*>(
+            &lds.vt[n * 16 + r][ds_ * 32 + qg * 8]);
+        bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+            &lds.qt[wid * 16 + r][ds_ * 32 + qg * 8]);
+        bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
+            &lds.dot[wid * 16 + r][ds_ * 32 + qg * 8]);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, bq, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bdo, dpt, 0, 0, 0);
+      }
+      const int q_glob = q0 + wid * 16 + r;
+      const float l_q = lds.lse_t[wid * 16 + r];
+      const float d_q = lds.delta_t[wid * 16 + r];
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int key_glob = k0 + n * 16 + qg * 4 + reg;
+        float p = 0.f;
+        if (!causal || q_glob >= key_glob)
+          p = __expf(scale * st[reg] - l_q);
+        ds_sub[n][reg] = p * (dpt[reg] - d_q) * scale;
+      }
+    }
+
+    // ---- route dS^T D-layout -> A[m=q][k=key 0..31] (bwd_route_probe)
+    unsigned pk[2][2];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      pk[n][0] = (unsigned)f2us(ds_sub[n][0]) |
+                 ((unsigned)f2us(ds_sub[n][1]) << 16);
+      pk[n][1] = (unsigned)f2us(ds_sub[n][2]) |
+                 ((unsigned)f2us(ds_sub[n][3]) << 16);
+    }
+    const int a = (qg & 1) * 2;   // first source quarter
+    const int nsrc = qg >> 1;     // source subtile (keys 0-15 / 16-31)
+    const int s0 = a * 16 + r;
+    const int s1 = (a + 1) * 16 + r;
+    // shuffle BOTH subtiles and select with the destination's nsrc after:
+    // __shfl evaluates its operand in the SOURCE lane, whose nsrc differs
+    unsigned rr[4][2];
+#pragma unroll
+    for (int nn = 0; nn < 2; ++nn) {
+      rr[nn * 2 + 0][0] = (unsigned)__shfl((int)pk[nn][0], s0, 64);
+      rr[nn * 2 + 0][1] = (unsigned)__shfl((int)pk[nn][1], s0, 64);
+      rr[nn * 2 + 1][0] = (unsigned)__shfl((int)pk[nn][0], s1, 64);
+      rr[nn * 2 + 1][1] = (unsigned)__shfl((int)pk[nn][1], s1, 64);
+    }
+    bf16x8 ads;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int src = (j >= 4) ? 1 : 0;
+      const int reg = j & 3;
+      const unsigned u = rr[nsrc * 2 + src][reg >> 1];
+      ads[j] = (short)((reg & 1) ? (u >> 16) : (u & 0xffffu));
+    }
+
+    // ---- dQ += dS @ K  (B-fragment: K transposed from the kt tile)
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+      bf16x8 bk;
+      if constexpr (TR16) {
+        bk = tr16_bfrag<D + 8>(&lds.kt[qg * 8][dt * 16], lane);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          bk[j] = (short)lds.kt[qg * 8 + j][dt * 16 + r];
+      }
+      acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ads, bk, acc[dt], 0, 0, 0);
+    }
+  }
+
+  // ---- store dQ once, bf16, [B,T,H,D]-physical strides
+  unsigned short* dQh = dQ16 + b * dqs.sb + h * dqs.sh;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int q_glob = q0 + wid * 16 + qg * 4 + reg;
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt)
+      dQh[(long)q_glob * dqs.st + dt * 16 + r] = f2us(acc[dt][reg]);
+  }
+}
+
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor o, at::Tensor lse,
                                  bool causal) {
@@ -919,9 +1083,9 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
 
   auto delta = at::empty({B, H, T}, q.options().dtype(at::kFloat));
   // grads physically [B, T, H, D] (matches the projection layout upstream,
-  // so the model-side transposes stay views)
-  auto dq_f32 = at::zeros({B, T, H, D}, q.options().dtype(at::kFloat))
-                    .permute({0, 2, 1, 3});
+  // so the model-side transposes stay views).  dQ is written exactly once
+  // by the q-block dQ kernel: bf16 direct, no fp32 scratch, no zero-fill.
+  auto dq = at::empty({B, T, H, D}, q.options()).permute({0, 2, 1, 3});
   const bool gqa = Hkv != H;
   auto dk = at::empty({B, T, Hkv, D}, k.options()).permute({0, 2, 1, 3});
   auto dv = at::empty({B, T, Hkv, D}, v.options()).permute({0, 2, 1, 3});
@@ -955,13 +1119,12 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     reinterpret_cast<const unsigned short*>(k.data_ptr()),                   \
     reinterpret_cast<const unsigned short*>(v.data_ptr()),                   \
     lse.data_ptr<float>(), delta.data_ptr<float>(),                          \
-    dq_f32.data_ptr<float>(),                                                \
     reinterpret_cast<unsigned short*>(dk.data_ptr()),                        \
     reinterpret_cast<unsigned short*>(dv.data_ptr()),                        \
     gqa ? dk_f32.data_ptr<float>() : nullptr,                                \
     gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,                     \
     scale, causal ? 1 : 0, str_of(dout), str_of(q),                          \
-    str_of(k), str_of(v), str_of(dq_f32),                                    \
+    str_of(k), str_of(v),                                                    \
     gqa ? str_of(dk_f32) : str_of(dk),                                       \
     gqa ? str_of(dv_f32) : str_of(dv)
 
@@ -974,6 +1137,42 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        reinterpret_cast<const unsigned short*>(o.data_ptr()), \
                        delta.data_ptr<float>(), T, H, str_of(dout),          \
                        str_of(o));                                           \
+    constexpr size_t SH_DQ = sizeof(AttnDqLds<DD>);                          \
+    static bool dq_attr_##DD = [] {                                          \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_dq_kernel<DD, false>),         \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_DQ);           \
+      hipFuncSetAttribute(                                                   \
+          reinterpret_cast<const void*>(&attn_dq_kernel<DD, true>),          \
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_DQ);           \
+      return true;                                                           \
+    }();                                                                     \
+    (void)dq_attr_##DD;                                                      \
+    {                                                                        \
+      dim3 qgrid(B * H, T / dq::QB), qblock(256);                            \
+      if (use_tr16 && DD <= 128)                                             \
+        hipLaunchKernelGGL((attn_dq_kernel<DD, true>), qgrid, qblock,        \
+                           SH_DQ, stream.stream(),                           \
+                           reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(q.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(k.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(v.data_ptr()), \
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),   \
+                           reinterpret_cast<unsigned short*>(dq.data_ptr()), \
+                           T, H, Hkv, scale, causal ? 1 : 0, str_of(dout),   \
+                           str_of(q), str_of(k), str_of(v), str_of(dq));     \
+      else                                                                   \
+        hipLaunchKernelGGL((attn_dq_kernel<DD, false>), qgrid, qblock,       \
+                           SH_DQ, stream.stream(),                           \
+                           reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(q.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(k.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(v.data_ptr()), \
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),   \
+                           reinterpret_cast<unsigned short*>(dq.data_ptr()), \
+                           T, H, Hkv, scale, causal ? 1 : 0, str_of(dout),   \
+                           str_of(q), str_of(k), str_of(v), str_of(dq));     \
+    }                                                                        \
     const bool tr = use_tr16 && DD <= 128;                                   \
     const bool vl2 = use_vl2 && DD <= 128;                                   \
     constexpr size_t SH_V = sizeof(AttnBwdLds<DD, (DD <= 128)>);             \
@@ -1013,9 +1212,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   else if (D == 128) LAUNCH_B(128);
   else LAUNCH_B(256);
 #undef LAUNCH_B
-  if (gqa) return {dq_f32.to(at::kBFloat16), dk_f32.to(at::kBFloat16),
-                   dv_f32.to(at::kBFloat16)};
-  return {dq_f32.to(at::kBFloat16), dk, dv};
+  if (gqa) return {dq, dk_f32.to(at::kBFloat16), dv_f32.to(at::kBFloat16)};
+  return {dq, dk, dv};
 }
 
 }  // namespace samd
