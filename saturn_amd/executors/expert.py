@@ -102,12 +102,18 @@ def _ep_worker(rank: int, world: int, task, tid: int, batch_count: int,
             step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
+                torch.cuda.reset_peak_memory_stats()
             t0 = timer()
             for _ in range(TRIAL_BATCHES - 1):
                 step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
-            return (timer() - t0) / (TRIAL_BATCHES - 1)
+            hbm = (
+                torch.cuda.max_memory_allocated() / 2**30
+                if device.type == "cuda"
+                else 0.0
+            )
+            return ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
         for _ in range(batch_count):
             step(next_batch())
         if device.type == "cuda":
@@ -147,12 +153,13 @@ class ExpertParallelExecutor(BaseTechnique):
         if len(gpus) < 2:
             return None, float("inf")  # EP needs >1 rank to shard experts
         try:
-            bt = gang_spawn(
+            out = gang_spawn(
                 _ep_worker, len(gpus), tid, task, tid, TRIAL_BATCHES,
                 {"ep": len(gpus)}, True,
             )
         except Exception:
             return None, float("inf")
-        if bt is None:
+        if out is None:
             return None, float("inf")
-        return {"ep": len(gpus)}, bt
+        bt, hbm = out
+        return dict(ep=len(gpus), hbm_peak_gb=round(hbm, 2)), bt

@@ -121,12 +121,18 @@ def _fsdp_worker(
             step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
+                torch.cuda.reset_peak_memory_stats()
             t0 = timer()
             for _ in range(TRIAL_BATCHES - 1):
                 step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
-            result = (timer() - t0) / (TRIAL_BATCHES - 1)
+            hbm = (
+                torch.cuda.max_memory_allocated() / 2**30
+                if device.type == "cuda"
+                else 0.0
+            )
+            result = ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
         else:
             for _ in range(batch_count):
                 step(next_batch())
@@ -179,11 +185,12 @@ class FSDPExecutor(BaseTechnique):
         world = len(gpus)
         for cfg in FSDPExecutor.GRID:
             try:
-                bt = gang_spawn(
+                out = gang_spawn(
                     _fsdp_worker, world, tid, task, tid, TRIAL_BATCHES, cfg, True
                 )
             except Exception:
                 continue
-            if bt is not None:
-                return dict(cfg), bt
+            if out is not None:
+                bt, hbm = out
+                return dict(cfg, hbm_peak_gb=round(hbm, 2)), bt
         return None, float("inf")

@@ -91,12 +91,18 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
+                torch.cuda.reset_peak_memory_stats()
             t0 = timer()
             for _ in range(TRIAL_BATCHES - 1):
                 step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
-            return (timer() - t0) / (TRIAL_BATCHES - 1)
+            hbm = (
+                torch.cuda.max_memory_allocated() / 2**30
+                if device.type == "cuda"
+                else 0.0
+            )
+            return ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
         RESYNC_EVERY = 64  # bound replicated-param drift from fp32 atomics
         for i in range(batch_count):
             step(next_batch())
@@ -139,12 +145,13 @@ class MegatronExecutor(BaseTechnique):
         if len(gpus) < 2:
             return None, float("inf")  # TP over 1 GPU is plain training
         try:
-            bt = gang_spawn(
+            out = gang_spawn(
                 _tp_worker, len(gpus), tid, task, tid, TRIAL_BATCHES,
                 {"tp": len(gpus)}, True,
             )
         except Exception:
             return None, float("inf")
-        if bt is None:
+        if out is None:
             return None, float("inf")
-        return {"tp": len(gpus)}, bt
+        bt, hbm = out
+        return dict(tp=len(gpus), hbm_peak_gb=round(hbm, 2)), bt

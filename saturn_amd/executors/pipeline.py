@@ -66,9 +66,25 @@ def _run_pipeline(task, n_devices: int, batch_count: int,
     model = task.get_model()
     model = model.to(dtype=dtype)
     seq = _to_sequential(task, model)
+    balance = params.get("balance")
+    if balance is None and params.get("time_balance", True):
+        # profile per-layer time on device 0 (reference balance_by_time,
+        # Pipeline.py:94-103); fall back to parameter balance if the whole
+        # model cannot visit one device
+        try:
+            from saturn_amd.parallel.pipeline import balance_by_time
+
+            x0, _ = next(task.get_fresh_iterator())
+            sample = x0[: max(1, x0.shape[0] // 4)]
+            if sample.is_floating_point():
+                sample = sample.to(dtype)
+            balance = balance_by_time(seq, sample, n_devices, devices[0])
+        except Exception:
+            balance = None
     pipe = PipelinedModel(
         seq,
         devices,
+        balance=balance,
         chunks=int(params.get("chunks", 4)),
         checkpoint_activations=bool(params.get("checkpoint", False)),
     )
@@ -105,11 +121,19 @@ def _run_pipeline(task, n_devices: int, batch_count: int,
     if trial:
         step(next_batch())
         sync()
+        if use_gpu:
+            for d in devices:
+                torch.cuda.reset_peak_memory_stats(d)
         t0 = timer()
         for _ in range(TRIAL_BATCHES - 1):
             step(next_batch())
         sync()
-        return (timer() - t0) / (TRIAL_BATCHES - 1)
+        hbm = (
+            max(torch.cuda.max_memory_allocated(d) for d in devices) / 2**30
+            if use_gpu
+            else 0.0
+        )
+        return ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
     for _ in range(batch_count):
         step(next_batch())
     sync()
@@ -150,10 +174,18 @@ class PipelineExecutor(BaseTechnique):
             chunks *= 2
         best: Tuple[Optional[Dict[str, Any]], float] = (None, float("inf"))
         for c in reversed(grid):  # most chunks first (reference halves down)
-            try:
-                bt = _run_pipeline(task, len(gpus), 0, {"chunks": c}, True)
-            except Exception:
-                continue
-            if bt < best[1]:
-                best = ({"chunks": c}, bt)
+            # activation-checkpoint knob: plain first, ckpt as the
+            # memory-rescue variant (first-fit, like the FSDP grid)
+            for ckpt in (False, True):
+                try:
+                    bt, hbm = _run_pipeline(
+                        task, len(gpus), 0,
+                        {"chunks": c, "checkpoint": ckpt}, True,
+                    )
+                except Exception:
+                    continue
+                if bt < best[1]:
+                    best = ({"chunks": c, "checkpoint": ckpt,
+                             "hbm_peak_gb": round(hbm, 2)}, bt)
+                break  # plain run fit -> no need to pay recompute
         return best

@@ -112,12 +112,16 @@ def _run_spilled(task, batch_count: int, params: Optional[Dict[str, Any]],
         step(next_batch())
         if use_gpu:
             torch.cuda.synchronize()
+            torch.cuda.reset_peak_memory_stats()
         t0 = timer()
         for _ in range(TRIAL_BATCHES - 1):
             step(next_batch())
         if use_gpu:
             torch.cuda.synchronize()
-        return (timer() - t0) / (TRIAL_BATCHES - 1)
+        hbm = (
+            torch.cuda.max_memory_allocated() / 2**30 if use_gpu else 0.0
+        )
+        return ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
     for _ in range(batch_count):
         step(next_batch())
     if use_gpu:
@@ -161,11 +165,11 @@ class SpilledExecutor(BaseTechnique):
         best: Tuple[Optional[Dict[str, Any]], float] = (None, float("inf"))
         for d in divisors[:4]:
             try:
-                bt = _run_spilled(task, 0, {"partitions": d}, True)
+                bt, hbm = _run_spilled(task, 0, {"partitions": d}, True)
             except Exception:
                 continue
             if bt < best[1]:
-                best = ({"partitions": d}, bt)
+                best = ({"partitions": d, "hbm_peak_gb": round(hbm, 2)}, bt)
             if best[0] is not None and d > 1:
                 break  # first fitting coarse partition wins (transfer-bound)
         return best

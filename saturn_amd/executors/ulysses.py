@@ -97,12 +97,18 @@ def _sp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
+                torch.cuda.reset_peak_memory_stats()
             t0 = timer()
             for _ in range(TRIAL_BATCHES - 1):
                 step(next_batch())
             if device.type == "cuda":
                 torch.cuda.synchronize()
-            result = (timer() - t0) / (TRIAL_BATCHES - 1)
+            hbm = (
+                torch.cuda.max_memory_allocated() / 2**30
+                if device.type == "cuda"
+                else 0.0
+            )
+            result = ((timer() - t0) / (TRIAL_BATCHES - 1), hbm)
         else:
             last = None
             for _ in range(batch_count):
@@ -146,12 +152,14 @@ class UlyssesExecutor(BaseTechnique):
         if len(gpus) < 2:
             return None, float("inf")  # SP over one GPU is plain training
         try:
-            bt = gang_spawn(
+            out = gang_spawn(
                 _sp_worker, len(gpus), tid, task, tid, TRIAL_BATCHES,
                 {"bucket_mb": 64.0}, True,
             )
         except Exception:
             return None, float("inf")
-        if bt is None:
+        if out is None:
             return None, float("inf")
-        return {"sp": len(gpus), "bucket_mb": 64.0}, bt
+        bt, hbm = out
+        return {"sp": len(gpus), "bucket_mb": 64.0,
+                "hbm_peak_gb": round(hbm, 2)}, bt

@@ -19,14 +19,7 @@ import torch
 import torch.nn as nn
 
 
-def balance_by_params(seq: nn.Sequential, n_stages: int) -> List[int]:
-    """Split layers into n_stages with ~equal parameter bytes (the reference
-    balances by profiled time, Pipeline.py:94-103; parameter balance is the
-    deterministic default and time balance is a search() refinement)."""
-    sizes = [
-        sum(p.numel() * p.element_size() for p in m.parameters()) + 1
-        for m in seq
-    ]
+def _greedy_partition(sizes: List[float], n_stages: int) -> List[int]:
     assert len(sizes) >= n_stages, "fewer layers than stages"
     balance: List[int] = []
     remaining = float(sum(sizes))
@@ -45,6 +38,60 @@ def balance_by_params(seq: nn.Sequential, n_stages: int) -> List[int]:
             stages_left -= 1
     balance.append(cnt)
     return balance
+
+
+def balance_by_params(seq: nn.Sequential, n_stages: int) -> List[int]:
+    """Split layers into n_stages with ~equal parameter bytes (deterministic
+    default; ``balance_by_time`` is the search() refinement)."""
+    sizes = [
+        sum(p.numel() * p.element_size() for p in m.parameters()) + 1.0
+        for m in seq
+    ]
+    return _greedy_partition(sizes, n_stages)
+
+
+def balance_by_time(
+    seq: nn.Sequential,
+    sample: torch.Tensor,
+    n_stages: int,
+    device: Optional[torch.device] = None,
+    reps: int = 3,
+) -> List[int]:
+    """Split layers by measured per-layer forward time (the reference's
+    torchgpipe ``balance_by_time``, Pipeline.py:94-103): wrong stage loads
+    under parameter balance when layers are uneven — GPT-J's embedding and
+    lm-head stages dwarf a block's bytes but not its time.
+
+    Profiles on ``device`` (whole model must fit one device; the caller
+    falls back to parameter balance on OOM).  The model is left on that
+    device — callers re-place stages afterwards.
+    """
+    from timeit import default_timer as timer
+
+    if device is None:
+        device = next(
+            (p.device for p in seq.parameters()), torch.device("cpu")
+        )
+    n = len(seq)
+    times = [0.0] * n
+
+    def sync() -> None:
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+
+    with torch.no_grad():
+        for rep in range(reps + 1):  # rep 0 = warmup (also moves layers)
+            h = sample.to(device)
+            for i, m in enumerate(seq):
+                if rep == 0:
+                    m.to(device)
+                sync()
+                t0 = timer()
+                h = m(h)
+                sync()
+                if rep > 0:
+                    times[i] += timer() - t0
+    return _greedy_partition([t + 1e-9 for t in times], n_stages)
 
 
 class PipelinedModel(nn.Module):

@@ -66,11 +66,16 @@ def search(
     n_gpus: Optional[int] = None,
     trial_timeout: Optional[float] = 1800.0,
     isolate: bool = True,
+    profile: bool = False,
 ) -> None:
     """Profile all cells and attach per-gpu-count winning Strategies.
 
     ``isolate=False`` runs cells in-process (CPU test mode — no HIP context
-    to isolate)."""
+    to isolate).  ``profile=True`` re-runs each task's fastest cell under
+    ``rocprofv3 --kernel-trace --stats`` and attaches the top kernel rows
+    plus a rocm-smi sample to that Strategy's parameters (the north star's
+    rocprof/rocm-smi-fed trial profiling; the reference used timeit only,
+    SURVEY §5.1)."""
     if log_level:
         logging.basicConfig(
             format="%(asctime)s %(levelname)-8s %(message)s",
@@ -173,3 +178,69 @@ def search(
                     runtime=bt * t.total_batches,
                     batch_time=bt,
                 )
+
+    if profile:
+        _profile_winners(tasks, n_gpus)
+
+
+def _profile_winners(tasks: List, n_gpus: int) -> None:
+    """Re-run each task's fastest feasible cell under rocprofv3 and attach
+    kernel-time + rocm-smi evidence to its Strategy parameters."""
+    import os
+    import tempfile
+
+    import dill
+
+    from saturn_amd.trial_runner.profiler import (
+        rocm_smi_sample,
+        rocprof_stats,
+    )
+
+    try:
+        import torch
+
+        has_gpu = torch.cuda.is_available()
+    except Exception:
+        has_gpu = False
+
+    for ti, t in enumerate(tasks):
+        feas = [
+            (s.batch_time, g, s)
+            for g, s in t.strategies.items()
+            if s is not None and s.feasible and s.batch_time is not None
+        ]
+        if not feas:
+            continue
+        _, g, strat = min(feas)
+        if strat.parameters is None:
+            continue
+        if not has_gpu:
+            continue  # rocprofv3 needs a device; CPU runs keep timeit-only
+        with tempfile.TemporaryDirectory() as td:
+            payload = os.path.join(td, "cell.pkl")
+            result = os.path.join(td, "out.pkl")
+            with open(payload, "wb") as fh:
+                dill.dump((strat.executor, t, g, 10_000 + ti), fh)
+            import sys as _sys
+
+            rows = rocprof_stats(
+                [_sys.executable, "-m", "saturn_amd.trial_runner.cell_main",
+                 payload, result]
+            )
+        if rows:
+            strat.parameters["kernels"] = [
+                {
+                    "name": (r.get("Name") or r.get("KernelName") or "?")[:100],
+                    "total_ns": r.get("TotalDurationNs")
+                    or r.get("DurationNs"),
+                    "calls": r.get("Calls") or r.get("TotalCalls"),
+                }
+                for r in rows[:10]
+            ]
+        smi = rocm_smi_sample()
+        if smi:
+            strat.parameters["rocm_smi"] = smi[: max(1, g)]
+        log.info(
+            "profiled winner for %s at %dG: %d kernel rows",
+            t.name, g, len(rows or []),
+        )

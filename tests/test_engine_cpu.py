@@ -202,3 +202,36 @@ def test_orchestrate_four_gpu_gang(save_dir, library_path):
     t.select_strategy(t.strategies[4])
     orchestrate([t], interval=120, n_gpus=4)
     assert t.has_ckpt()
+
+
+def test_search_profile_attaches_kernel_evidence(save_dir, library_path, monkeypatch):
+    """profile=True must attach rocprof kernel rows + rocm-smi samples to
+    the fastest feasible Strategy's params (north-star rocprof-fed trials;
+    mocked here — the real rocprofv3 path needs a GPU)."""
+    import torch
+
+    import saturn_amd.trial_runner.profiler as prof
+
+    register("ddp", DDPExecutor)
+    t = make_mlp_task("prof_a", save_dir)
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(
+        prof, "rocprof_stats",
+        lambda cmd, **kw: [
+            {"Name": "samd::fused_sgd_kernel", "TotalDurationNs": "1200",
+             "Calls": "4"},
+            {"Name": "Cijk_gemm", "TotalDurationNs": "900", "Calls": "12"},
+        ],
+    )
+    monkeypatch.setattr(
+        prof, "rocm_smi_sample",
+        lambda: [{"card": "card0", "VRAM Total Used Memory (B)": "123"}],
+    )
+    search([t], log_level=False, n_gpus=2, isolate=False, profile=True)
+    best = min(
+        (s for s in t.strategies.values() if s.feasible),
+        key=lambda s: s.batch_time,
+    )
+    assert "kernels" in best.parameters, best.parameters
+    assert best.parameters["kernels"][0]["name"].startswith("samd::")
+    assert "rocm_smi" in best.parameters

@@ -194,3 +194,31 @@ def test_pipeline_balance_covers_all_layers(sizes, n_stages):
     assert len(bal) == n_stages
     assert all(b >= 1 for b in bal), bal
     assert sum(bal) == len(seq), (bal, len(seq))
+
+
+def test_balance_by_time_vs_params():
+    """Time balance must react to uneven layer cost where parameter
+    balance cannot (reference balance_by_time, Pipeline.py:94-103)."""
+    import torch
+    import torch.nn as nn
+
+    from saturn_amd.parallel.pipeline import (
+        balance_by_params,
+        balance_by_time,
+    )
+
+    class _Rep(nn.Module):
+        def __init__(self, reps):
+            super().__init__()
+            self.reps = reps
+
+        def forward(self, x):
+            for _ in range(self.reps):
+                x = torch.tanh(x)
+            return x
+
+    seq = nn.Sequential(_Rep(400), _Rep(1), _Rep(1), _Rep(1))
+    assert balance_by_params(seq, 2) == [2, 2]  # no params -> naive split
+    bal = balance_by_time(seq, torch.randn(256, 256), 2)
+    assert sum(bal) == 4 and len(bal) == 2
+    assert bal[0] == 1, bal  # the slow layer dominates stage 0
