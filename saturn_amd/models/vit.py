@@ -2,10 +2,10 @@
 config 4).
 
 Standard pre-LN ViT: patch embed (conv), cls token, learned positions,
-bidirectional attention.  Sequence length 197 is not a multiple of 64, so
-attention rides the explicit-GEMM math path (the flash kernel's shape guard
-dispatches automatically); classification loss is plain CE over 1000
-classes.
+bidirectional attention.  Sequence length 197 is not a multiple of 64;
+the flash wrapper zero-pads to 256 and masks the padded keys in-kernel
+(ragged-T support), so ViT rides the fused MFMA attention path too.
+Classification loss is plain CE over 1000 classes.
 """
 
 from __future__ import annotations

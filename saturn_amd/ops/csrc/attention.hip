@@ -130,8 +130,9 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
                                 const unsigned short* __restrict__ V,
                                 unsigned short* __restrict__ O,
                                 float* __restrict__ LSE, int T, int n_heads,
-                                int n_kv, float scale, int causal, TStr qs,
-                                TStr ks, TStr vs, TStr os) {
+                                int n_kv, float scale, int causal,
+                                int kv_len, TStr qs, TStr ks, TStr vs,
+                                TStr os) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnLds<D>& lds = *reinterpret_cast<AttnLds<D>*>(smem);
 
@@ -229,6 +230,9 @@ __global__ void attn_fwd_kernel(const unsigned short* __restrict__ Q,
         if (kv0 + r > qrow) s0 = -INFINITY;
         if (kv0 + 16 + r > qrow) s1 = -INFINITY;
       }
+      // ragged T: keys beyond the unpadded length contribute nothing
+      if (kv0 + r >= kv_len) s0 = -INFINITY;
+      if (kv0 + 16 + r >= kv_len) s1 = -INFINITY;
       float tm = fmaxf(s0, s1);
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
@@ -307,7 +311,8 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
                                    unsigned short* __restrict__ O,
                                    float* __restrict__ LSE, int T,
                                    int n_heads, int n_kv, float scale,
-                                   int causal, TStr qs, TStr ks, TStr vs,
+                                   int causal, int kv_len, TStr qs,
+                                   TStr ks, TStr vs,
                                    TStr os) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnV2Lds<D>& lds = *reinterpret_cast<AttnV2Lds<D>*>(smem);
@@ -407,7 +412,8 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
     for (int reg = 0; reg < 16; ++reg) {
       const int key_loc = (reg & 3) + 8 * (reg >> 2) + 4 * hi;
       float sv = st[reg] * scale;
-      if (causal && kv0 + key_loc > q_glob) sv = -INFINITY;
+      if ((causal && kv0 + key_loc > q_glob) || kv0 + key_loc >= kv_len)
+        sv = -INFINITY;
       p[reg] = sv;
       tile_max = fmaxf(tile_max, sv);
     }
@@ -524,7 +530,9 @@ __global__ void attn_fwd_v2_kernel(const unsigned short* __restrict__ Q,
 // Host wrapper
 // ---------------------------------------------------------------------------
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
-                                 bool causal) {
+                                 bool causal, int64_t kv_len) {
+  // kv_len < T marks ragged sequences: q/k/v are zero-padded to a 64
+  // multiple by the wrapper and keys >= kv_len are masked out in-kernel
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16,
               "attn_fwd: bf16 only");
   TORCH_CHECK(q.dim() == 4, "attn_fwd: [B, H, T, D]");
@@ -536,6 +544,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(H % Hkv == 0, "attn_fwd: q heads must be a multiple of kv heads");
   TORCH_CHECK(k.size(2) == T, "attn_fwd: q/k length mismatch");
   TORCH_CHECK(T % QBLK == 0, "attn_fwd: T must be a multiple of 64");
+  if (kv_len <= 0) kv_len = T;
+  TORCH_CHECK(kv_len <= T, "attn_fwd: kv_len exceeds padded T");
   TORCH_CHECK(D == 64 || D == 128 || D == 256, "attn_fwd: D in {64,128,256}");
   TORCH_CHECK((reinterpret_cast<uintptr_t>(q.data_ptr()) & 15) == 0 &&
               q.stride(2) % 8 == 0 && k.stride(2) % 8 == 0 &&
@@ -569,8 +579,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                          reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                          reinterpret_cast<unsigned short*>(o.data_ptr()),    \
                          lse.data_ptr<float>(), T, H, Hkv, scale,            \
-                         causal ? 1 : 0, str_of(q), str_of(k), str_of(v),    \
-                         str_of(o));                                         \
+                         causal ? 1 : 0, (int)kv_len, str_of(q), str_of(k),  \
+                         str_of(v), str_of(o));                              \
     } while (0)
     if (D == 64) LAUNCH_V2(64);
     else LAUNCH_V2(128);
@@ -589,8 +599,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                        reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                        reinterpret_cast<unsigned short*>(o.data_ptr()),     \
                        lse.data_ptr<float>(), T, H, Hkv, scale,             \
-                       causal ? 1 : 0, str_of(q), str_of(k), str_of(v),     \
-                       str_of(o));                                          \
+                       causal ? 1 : 0, (int)kv_len, str_of(q), str_of(k),   \
+                       str_of(v), str_of(o));                               \
   } while (0)
 
   if (D == 64) LAUNCH(64);
@@ -690,7 +700,8 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
                                 unsigned short* __restrict__ dV,
                                 float* __restrict__ dKf,
                                 float* __restrict__ dVf, int T, int n_heads,
-                                int n_kv, float scale, int causal, TStr dos,
+                                int n_kv, float scale, int causal,
+                                int kv_len, TStr dos,
                                 TStr qs, TStr ks, TStr vs,
                                 TStr dks, TStr dvs) {
   using namespace bwd;
@@ -791,7 +802,7 @@ __global__ void attn_bwd_kernel(const unsigned short* __restrict__ dO,
         const int key_glob = kv0 + wid * 16 + key_loc;
         const int q_glob = q0 + n * 16 + r;
         float p = 0.f;
-        if (!causal || q_glob >= key_glob) {
+        if ((!causal || q_glob >= key_glob) && key_glob < kv_len) {
           p = __expf(scale * st[reg] - lds.lse_t[n * 16 + r]);
         }
         const float dsv = p * (dpt[reg] - lds.delta_t[n * 16 + r]) * scale;
@@ -908,8 +919,8 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
                                const float* __restrict__ DELTA,
                                unsigned short* __restrict__ dQ16, int T,
                                int n_heads, int n_kv, float scale, int causal,
-                               TStr dos, TStr qs, TStr ks, TStr vs,
-                               TStr dqs) {
+                               int kv_len, TStr dos, TStr qs, TStr ks,
+                               TStr vs, TStr dqs) {
   using namespace dq;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   AttnDqLds<D>& lds = *reinterpret_cast<AttnDqLds<D>*>(smem);
@@ -1001,7 +1012,7 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
       for (int reg = 0; reg < 4; ++reg) {
         const int key_glob = k0 + n * 16 + qg * 4 + reg;
         float p = 0.f;
-        if (!causal || q_glob >= key_glob)
+        if ((!causal || q_glob >= key_glob) && key_glob < kv_len)
           p = __expf(scale * st[reg] - l_q);
         ds_sub[n][reg] = p * (dpt[reg] - d_q) * scale;
       }
@@ -1067,7 +1078,7 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
 
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor o, at::Tensor lse,
-                                 bool causal) {
+                                 bool causal, int64_t kv_len) {
   using namespace bwd;
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
   const int B = (int)q.size(0), H = (int)q.size(1), T = (int)q.size(2),
@@ -1075,6 +1086,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
   const int Hkv = (int)k.size(1);
   TORCH_CHECK(H % Hkv == 0);
   TORCH_CHECK(T % KB == 0, "attn_bwd: T must be a multiple of 64");
+  if (kv_len <= 0) kv_len = T;
+  TORCH_CHECK(kv_len <= T, "attn_bwd: kv_len exceeds padded T");
   TORCH_CHECK(D == 64 || D == 128 || D == 256);
   TORCH_CHECK(dout.stride(3) == 1 && q.stride(3) == 1 && k.stride(3) == 1 &&
               v.stride(3) == 1, "attn_bwd: innermost dim must be dense");
@@ -1123,7 +1136,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     reinterpret_cast<unsigned short*>(dv.data_ptr()),                        \
     gqa ? dk_f32.data_ptr<float>() : nullptr,                                \
     gqa ? dv_f32.data_ptr<float>() : nullptr, T, H, Hkv,                     \
-    scale, causal ? 1 : 0, str_of(dout), str_of(q),                          \
+    scale, causal ? 1 : 0, (int)kv_len, str_of(dout), str_of(q),             \
     str_of(k), str_of(v),                                                    \
     gqa ? str_of(dk_f32) : str_of(dk),                                       \
     gqa ? str_of(dv_f32) : str_of(dv)
@@ -1159,8 +1172,9 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                            reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                            lse.data_ptr<float>(), delta.data_ptr<float>(),   \
                            reinterpret_cast<unsigned short*>(dq.data_ptr()), \
-                           T, H, Hkv, scale, causal ? 1 : 0, str_of(dout),   \
-                           str_of(q), str_of(k), str_of(v), str_of(dq));     \
+                           T, H, Hkv, scale, causal ? 1 : 0, (int)kv_len,   \
+                           str_of(dout), str_of(q), str_of(k), str_of(v),    \
+                           str_of(dq));                                      \
       else                                                                   \
         hipLaunchKernelGGL((attn_dq_kernel<DD, false>), qgrid, qblock,       \
                            SH_DQ, stream.stream(),                           \
@@ -1170,8 +1184,9 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                            reinterpret_cast<const unsigned short*>(v.data_ptr()), \
                            lse.data_ptr<float>(), delta.data_ptr<float>(),   \
                            reinterpret_cast<unsigned short*>(dq.data_ptr()), \
-                           T, H, Hkv, scale, causal ? 1 : 0, str_of(dout),   \
-                           str_of(q), str_of(k), str_of(v), str_of(dq));     \
+                           T, H, Hkv, scale, causal ? 1 : 0, (int)kv_len,   \
+                           str_of(dout), str_of(q), str_of(k), str_of(v),    \
+                           str_of(dq));                                      \
     }                                                                        \
     const bool tr = use_tr16 && DD <= 128;                                   \
     const bool vl2 = use_vl2 && DD <= 128;                                   \

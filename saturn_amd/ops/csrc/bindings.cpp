@@ -27,10 +27,10 @@ void rope_apply(at::Tensor y, at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
 at::Tensor gelu_fwd(at::Tensor x);
 at::Tensor gelu_bwd(at::Tensor dy, at::Tensor x);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
-                                 bool causal);
+                                 bool causal, int64_t kv_len);
 std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                                  at::Tensor v, at::Tensor o, at::Tensor lse,
-                                 bool causal);
+                                 bool causal, int64_t kv_len);
 at::Tensor add3(at::Tensor a, at::Tensor b, at::Tensor c);
 at::Tensor swiglu_fwd(at::Tensor gate, at::Tensor up);
 std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor gate,
@@ -53,8 +53,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused rotary embedding (out-of-place, offset table rows)");
   m.def("gelu_fwd", &samd::gelu_fwd, "fused tanh-approx GELU forward");
   m.def("gelu_bwd", &samd::gelu_bwd, "fused tanh-approx GELU backward");
-  m.def("attn_fwd", &samd::attn_fwd, "fused causal flash attention forward");
-  m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward");
+  m.def("attn_fwd", &samd::attn_fwd, "fused flash attention forward",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
+        py::arg("kv_len") = -1);
+  m.def("attn_bwd", &samd::attn_bwd, "fused flash attention backward",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("causal"),
+        py::arg("kv_len") = -1);
   m.def("add3", &samd::add3, "fused 3-way residual add");
   m.def("swiglu_fwd", &samd::swiglu_fwd, "fused silu(gate)*up");
   m.def("swiglu_bwd", &samd::swiglu_bwd, "fused SwiGLU backward");

@@ -19,23 +19,41 @@ log = logging.getLogger(__name__)
 _warned = False
 
 
+def _pad_t64(t: torch.Tensor) -> torch.Tensor:
+    """Zero-pad the T dim (dim -2) up to a multiple of 64."""
+    pad = (-t.shape[-2]) % 64
+    if pad == 0:
+        return t
+    return torch.nn.functional.pad(t, (0, 0, 0, pad))
+
+
 class _FlashFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         ext = require_ext()
-        o, lse = ext.attn_fwd(q, k, v, causal)
+        T = q.shape[-2]
+        if T % 64:
+            # ragged T (ViT's 197, BERT's raw lengths): zero-pad to the
+            # kernels' 64-row tiles; padded KEYS are masked in-kernel via
+            # kv_len and padded q rows are sliced off below (their zero
+            # dO rows contribute exactly 0 to dK/dV in backward)
+            q, k, v = _pad_t64(q), _pad_t64(k), _pad_t64(v)
+        o, lse = ext.attn_fwd(q, k, v, causal, T)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
-        return o
+        ctx.t_real = T
+        return o[:, :, :T] if T % 64 else o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         ext = require_ext()
+        T = ctx.t_real
         if hasattr(ext, "attn_bwd"):
-            dq, dk, dv = ext.attn_bwd(
-                _dense_rows(do), q, k, v, o, lse, ctx.causal
-            )
+            do = _pad_t64(_dense_rows(do)) if T % 64 else _dense_rows(do)
+            dq, dk, dv = ext.attn_bwd(do, q, k, v, o, lse, ctx.causal, T)
+            if T % 64:
+                dq, dk, dv = dq[:, :, :T], dk[:, :, :T], dv[:, :, :T]
             return dq, dk, dv, None
         # Analytic FA2 backward composed from rocBLAS GEMMs (recompute P
         # from the saved LSE; O(T^2) transient, fp32 math).  The fused HIP
@@ -59,11 +77,9 @@ class _FlashFn(torch.autograd.Function):
 
 
 def _kernel_supported(q) -> bool:
-    return (
-        q.dtype == torch.bfloat16
-        and q.shape[-2] % 64 == 0
-        and q.shape[-1] in (64, 128, 256)
-    )
+    # ragged T is padded to 64 rows by the wrapper; D stays restricted to
+    # the MFMA tile shapes
+    return q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128, 256)
 
 
 def _dense_rows(t):

@@ -552,3 +552,48 @@ def test_dropout_p0_and_eval_identity():
     assert fused_dropout(x, 0.0) is x
     d = FusedDropout(0.5).eval()
     assert d(x) is x
+
+
+# ---------------------------------------------------------------------------
+# Ragged-T attention (kv_len masking; ViT's T=197 rides the fused path)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("T,causal", [(197, False), (100, False), (197, True)])
+def test_flash_attention_ragged_T(T, causal):
+    requires_ext()
+    from saturn_amd.ops.flash import flash_attention
+    from saturn_amd.ops.functional import attention_math
+
+    torch.manual_seed(0)
+    B, H, D = 2, 4, 64
+    q32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    k32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    v32 = torch.randn(B, H, T, D, device="cuda", requires_grad=True)
+    ref = attention_math(q32, k32, v32, causal=causal)
+    do = torch.randn_like(ref)
+    ref.backward(do)
+
+    q = q32.detach().to(torch.bfloat16).requires_grad_(True)
+    k = k32.detach().to(torch.bfloat16).requires_grad_(True)
+    v = v32.detach().to(torch.bfloat16).requires_grad_(True)
+    out = flash_attention(q, k, v, causal=causal)
+    assert out.shape == (B, H, T, D)
+    assert rel_err(out, ref) < 3e-2
+    out.backward(do.to(torch.bfloat16))
+    assert rel_err(q.grad, q32.grad) < 5e-2
+    assert rel_err(k.grad, k32.grad) < 5e-2
+    assert rel_err(v.grad, v32.grad) < 5e-2
+
+
+def test_vit_forward_backward_fused():
+    """ViT-L's T=197 runs the fused attention via padding (BASELINE
+    config 4's heterogeneous batch member)."""
+    requires_ext()
+    from saturn_amd.models.vit import get_vit_model, vit_loss
+
+    torch.manual_seed(0)
+    m = get_vit_model({"n_layer": 2}).to("cuda", torch.bfloat16)
+    x = torch.randn(2, 3, 224, 224, device="cuda", dtype=torch.bfloat16)
+    y = torch.randint(0, 1000, (2,), device="cuda")
+    loss = vit_loss(m(x), y)
+    loss.backward()
+    assert torch.isfinite(loss.detach())
