@@ -62,7 +62,10 @@ def _ddp_worker(
         device = torch.device("cuda", rank) if backend == "nccl" else torch.device("cpu")
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
 
-        model = task.get_model()
+        # build directly on the device: host-side init + H2D copy of a
+        # 6B-class model dominates a short trial cell
+        with device:
+            model = task.get_model()
         model = model.to(device=device, dtype=dtype)
         model.train()
         bucket_mb = float((params or {}).get("bucket_mb", 64.0))

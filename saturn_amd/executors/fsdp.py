@@ -46,7 +46,8 @@ def _fsdp_worker(
         )
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
 
-        model = task.get_model()
+        with device:  # device-side build (see ddp.py)
+            model = task.get_model()
         model = model.to(device=device, dtype=dtype)
         model.train()
         z3 = Zero3Model(

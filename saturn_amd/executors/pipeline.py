@@ -63,7 +63,8 @@ def _run_pipeline(task, n_devices: int, batch_count: int,
     )
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
-    model = task.get_model()
+    with devices[0]:  # device-side build (see ddp.py)
+        model = task.get_model()
     model = model.to(dtype=dtype)
     seq = _to_sequential(task, model)
     balance = params.get("balance")

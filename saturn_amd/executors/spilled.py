@@ -64,7 +64,9 @@ def _run_spilled(task, batch_count: int, params: Optional[Dict[str, Any]],
     device = torch.device("cuda", 0) if use_gpu else torch.device("cpu")
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
-    model = task.get_model().to(device=device, dtype=dtype)
+    with device:  # device-side build (see ddp.py)
+        model = task.get_model()
+    model = model.to(device=device, dtype=dtype)
     model.train()
     blocks = _blocks_of(task, model)
     n_part = int(params.get("partitions", len(blocks)))
