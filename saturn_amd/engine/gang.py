@@ -177,10 +177,22 @@ def execute(
                 visible,
                 batches_to_run[r_idx],
             )
+            # Per-gang host-CPU reservation (reference executor.py:107
+            # reserved num_cpus = g*3/4 per gang via Ray): cap the BLAS/
+            # OMP thread pools to this gang's proportional share so
+            # concurrent gangs don't oversubscribe host cores during
+            # dataloading/checkpointing.
+            n_vis = max(1, len(gpus))
+            total_gpus = max(n_vis, len({g for s in plan.gpu_sets for g in s}))
+            cpu_share = max(
+                1, (os.cpu_count() or 8) * 3 // 4 * n_vis // total_gpus
+            )
             env = {
                 "HIP_VISIBLE_DEVICES": visible,
                 "CUDA_VISIBLE_DEVICES": visible,
                 "SATURN_TASK_ID": str(plan_idx[task.name]),
+                "OMP_NUM_THREADS": str(cpu_share),
+                "MKL_NUM_THREADS": str(cpu_share),
             }
             run_in_subprocess(
                 _launch_task,

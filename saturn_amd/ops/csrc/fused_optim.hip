@@ -231,8 +231,11 @@ __global__ void fused_adam_kernel(AdamArgs a) {
 }
 
 static int grid_for(long total, int block) {
+  // HBM-bound sweep: more resident blocks = more outstanding loads (the
+  // 2048 cap measured 4.9 TB/s on the 6B update; 4096 keeps every CU at
+  // 2 blocks with a 16-deep grid-stride)
   long g = (total + block - 1) / block;
-  return (int)std::min<long>(std::max<long>(g, 1), 2048);
+  return (int)std::min<long>(std::max<long>(g, 1), 4096);
 }
 
 static long groups_of(long numel) { return (numel + GRP - 1) / GRP; }
