@@ -28,6 +28,31 @@ from saturn_amd.executors.launch import (
 TRIAL_BATCHES = 3  # 1 warmup + 2 timed (reference used 2: DDP.py:43)
 
 
+def build_model_on(task, device, dtype):
+    """Build the task's model directly on ``device`` in ``dtype``.
+
+    Host-side fp32 init + H2D copy of a large model dominates short trial
+    cells, and a 70B-class model can't even transit fp32-on-device
+    (280 GB weights + the bf16 copy during .to()).  Building under the
+    device context with the default dtype set to the compute dtype
+    materializes weights once, on-device, at final precision.
+    """
+    import torch
+
+    if device.type != "cuda":
+        return task.get_model().to(device=device, dtype=dtype)
+    prev = torch.get_default_dtype()
+    torch.set_default_dtype(dtype)
+    try:
+        with device:
+            model = task.get_model()
+    finally:
+        torch.set_default_dtype(prev)
+    # .to() is a no-op for params already in (device, dtype); it still
+    # sweeps stray buffers a factory may have made elsewhere
+    return model.to(device=device, dtype=dtype)
+
+
 def _make_optimizer(task, model):
     """Task-specified optimizer, defaulting to the framework's fused SGD on
     GPU / torch SGD on CPU."""
@@ -62,11 +87,7 @@ def _ddp_worker(
         device = torch.device("cuda", rank) if backend == "nccl" else torch.device("cpu")
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
 
-        # build directly on the device: host-side init + H2D copy of a
-        # 6B-class model dominates a short trial cell
-        with device:
-            model = task.get_model()
-        model = model.to(device=device, dtype=dtype)
+        model = build_model_on(task, device, dtype)
         model.train()
         bucket_mb = float((params or {}).get("bucket_mb", 64.0))
         comm = None

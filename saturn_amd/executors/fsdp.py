@@ -46,9 +46,9 @@ def _fsdp_worker(
         )
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
 
-        with device:  # device-side build (see ddp.py)
-            model = task.get_model()
-        model = model.to(device=device, dtype=dtype)
+        from saturn_amd.executors.ddp import build_model_on
+
+        model = build_model_on(task, device, dtype)
         model.train()
         z3 = Zero3Model(
             model,

@@ -40,9 +40,9 @@ def _tp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             torch.device("cuda", rank) if backend == "nccl" else torch.device("cpu")
         )
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
-        with device:  # device-side build (see ddp.py)
-            model = task.get_model()
-        model = model.to(device=device, dtype=dtype)
+        from saturn_amd.executors.ddp import build_model_on
+
+        model = build_model_on(task, device, dtype)
         model.train()
         if world > 1:
             with torch.no_grad():

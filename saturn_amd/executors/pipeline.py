@@ -63,9 +63,9 @@ def _run_pipeline(task, n_devices: int, batch_count: int,
     )
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
-    with devices[0]:  # device-side build (see ddp.py)
-        model = task.get_model()
-    model = model.to(dtype=dtype)
+    from saturn_amd.executors.ddp import build_model_on
+
+    model = build_model_on(task, devices[0], dtype)
     seq = _to_sequential(task, model)
     balance = params.get("balance")
     if balance is None and params.get("time_balance", True):

@@ -64,9 +64,9 @@ def _run_spilled(task, batch_count: int, params: Optional[Dict[str, Any]],
     device = torch.device("cuda", 0) if use_gpu else torch.device("cpu")
     dtype = torch.bfloat16 if use_gpu else torch.float32
 
-    with device:  # device-side build (see ddp.py)
-        model = task.get_model()
-    model = model.to(device=device, dtype=dtype)
+    from saturn_amd.executors.ddp import build_model_on
+
+    model = build_model_on(task, device, dtype)
     model.train()
     blocks = _blocks_of(task, model)
     n_part = int(params.get("partitions", len(blocks)))

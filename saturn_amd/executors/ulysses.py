@@ -48,9 +48,9 @@ def _sp_worker(rank: int, world: int, task, tid: int, batch_count: int,
             torch.device("cuda", rank) if backend == "nccl" else torch.device("cpu")
         )
         dtype = torch.bfloat16 if backend == "nccl" else torch.float32
-        with device:  # device-side build (see ddp.py)
-            model = task.get_model()
-        model = model.to(device=device, dtype=dtype)
+        from saturn_amd.executors.ddp import build_model_on
+
+        model = build_model_on(task, device, dtype)
         model.train()
         ddp = BucketedDDP(model, bucket_mb=float((params or {}).get("bucket_mb", 64.0)))
         optimizer = _make_optimizer(task, model)
