@@ -139,7 +139,16 @@ def main() -> None:
     ap.add_argument("--scale", choices=["tiny", "full"], default="tiny")
     ap.add_argument("--n-gpus", type=int, default=None)
     ap.add_argument("--interval", type=float, default=None)
+    ap.add_argument(
+        "--ckpt", choices=["all", "none"], default="all",
+        help="'none' sets SATURN_SKIP_CKPT=1: measure makespan without "
+        "checkpoint IO (for boxes whose scratch disk cannot hold the "
+        "batch's model artifacts); interval migration then restarts "
+        "jobs from init, so use it only for timing runs",
+    )
     args = ap.parse_args()
+    if args.ckpt == "none":
+        os.environ["SATURN_SKIP_CKPT"] = "1" 
 
     import torch
 
@@ -215,6 +224,7 @@ def main() -> None:
             "n_jobs": len(tasks),
             "techniques": execs,
         },
+        "ckpt": args.ckpt,
         "search_time_s": round(t_search, 2),
         "predicted_makespan_s": round(plan.makespan, 2),
         "plan": plan_dump,

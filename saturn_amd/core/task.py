@@ -195,6 +195,13 @@ class Task:
         optimizer moments at every interval boundary; we keep
         both under the same ``<name>.pt`` path.
         """
+        if os.environ.get("SATURN_SKIP_CKPT") == "1":
+            # Opt-in for pure-makespan benchmarking on scratch disks that
+            # cannot hold the batch's model artifacts (a full-scale 8-job
+            # Llama-8B sweep writes 8 x 16 GB; measured ENOSPC on the GPU
+            # pool's boxes).  Durability/migration semantics are unchanged
+            # when unset — this is never set by the orchestrator itself.
+            return
         state = model if isinstance(model, dict) else model.state_dict()
         payload = {
             "model": {k: v.cpu() for k, v in state.items()},
@@ -204,6 +211,20 @@ class Task:
         tmp = self.ckpt_path + ".tmp"
         torch.save(payload, tmp)
         os.replace(tmp, self.ckpt_path)  # atomic: no torn ckpt on crash
+
+    def delete_checkpoint(self) -> None:
+        """Remove the checkpoint and any per-rank optimizer-shard files —
+        called by the orchestrator when the task completes (disk hygiene
+        for large batches)."""
+        import glob
+
+        for p in [self.ckpt_path] + glob.glob(
+            os.path.join(self.save_dir, f"{self.name}.*.pt")
+        ):
+            try:
+                os.remove(p)
+            except OSError:
+                pass
 
     def load_checkpoint(self) -> Optional[Dict[str, Any]]:
         if not self.has_ckpt():

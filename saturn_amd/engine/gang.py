@@ -124,6 +124,7 @@ def execute(
     task_dependency_dict: Optional[Dict] = None,
     launch_timeout: Optional[float] = None,
     raise_on_failure: bool = True,
+    skip_ckpt_names: Optional[set] = None,
 ) -> List:
     """Execute one interval's tasks with MILP-ordered gang placement.
 
@@ -194,6 +195,11 @@ def execute(
                 "OMP_NUM_THREADS": str(cpu_share),
                 "MKL_NUM_THREADS": str(cpu_share),
             }
+            if skip_ckpt_names and task.name in skip_ckpt_names:
+                # task completes inside this interval: no resume will ever
+                # read its checkpoint, so skip the (potentially tens of
+                # GB) write — disk hygiene for full-scale batches
+                env["SATURN_SKIP_CKPT"] = "1" 
             run_in_subprocess(
                 _launch_task,
                 task,

@@ -123,9 +123,10 @@ def _ep_worker(rank: int, world: int, task, tid: int, batch_count: int,
         sd = ep_full_state_dict(model)
         if rank == 0 and sd is not None:
             task.save_checkpoint(sd, None)
-        tmp = opt_path + ".tmp"
-        torch.save(optimizer.state_dict(), tmp)
-        _os.replace(tmp, opt_path)
+        if _os.environ.get("SATURN_SKIP_CKPT") != "1":
+            tmp = opt_path + ".tmp"
+            torch.save(optimizer.state_dict(), tmp)
+            _os.replace(tmp, opt_path)
         if world > 1:
             dist.barrier()
         return None

@@ -142,10 +142,12 @@ def _fsdp_worker(
             sd = z3.full_state_dict()
             if rank == 0 and sd is not None:
                 task.save_checkpoint(sd, None)
-            # every rank persists its optimizer shard state
-            tmp = opt_path + ".tmp"
-            torch.save(optimizer.state_dict(), tmp)
-            _os.replace(tmp, opt_path)
+            # every rank persists its optimizer shard state (skipped
+            # when the engine marked this task completing — see Task)
+            if _os.environ.get("SATURN_SKIP_CKPT") != "1":
+                tmp = opt_path + ".tmp"
+                torch.save(optimizer.state_dict(), tmp)
+                _os.replace(tmp, opt_path)
             import torch.distributed as dist
 
             if world > 1:
