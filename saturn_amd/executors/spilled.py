@@ -155,6 +155,28 @@ class SpilledExecutor(BaseTechnique):
     ) -> Tuple[Optional[Dict[str, Any]], float]:
         if len(gpus) != 1:
             return None, float("inf")
+        # Spilling exists for models that DON'T fit HBM: layer streaming is
+        # strictly slower than resident training when everything fits, so
+        # running the divisor-grid trials for a comfortably-fitting model
+        # only burns search time (round-2 config-4: 290 s of search for a
+        # 48 s makespan, mostly spill trials the solver never picked).
+        # A meta-device build counts parameters for free.
+        try:
+            import torch
+
+            if torch.cuda.is_available():
+                with torch.device("meta"):
+                    meta_model = task.get_model(fresh=True)
+                pbytes = sum(
+                    p.numel() * 2 for p in meta_model.parameters()  # bf16
+                )
+                del meta_model
+                _free, total = torch.cuda.mem_get_info()
+                # weights + grads + optimizer + activation slack ~4x
+                if pbytes * 4 < total * 0.7:
+                    return None, float("inf")  # resident techniques win
+        except Exception:
+            pass  # meta build unsupported -> fall through to real trials
         # probe block count for the divisor grid (reference Spilled.py:91-96)
         try:
             model = task.get_model(fresh=True)
