@@ -76,8 +76,20 @@ class Plan:
 
     def shift(self, seconds: float) -> None:
         """Advance the plan by one executed interval (reference
-        milp.py:434-442)."""
-        self.start_times = [max(0.0, s - seconds) for s in self.start_times]
+        milp.py:434-442).
+
+        A task that was RUNNING during the shifted window has executed
+        ``min(seconds - start, runtime)`` of its work, so its remaining
+        runtime shrinks along with the clamped start — otherwise the kept
+        incumbent shows phantom overlaps between a task whose start
+        clamped to 0 and its successors (fuzz-caught in round 2; the
+        reference avoids it only because it destructively decrements
+        strategy runtimes elsewhere, executor.py:166-172)."""
+        for i in range(len(self.task_names)):
+            s, r = self.start_times[i], self.runtimes[i]
+            executed = min(max(seconds - s, 0.0), r)
+            self.start_times[i] = max(0.0, s - seconds)
+            self.runtimes[i] = r - executed
         self.makespan = max(0.0, self.makespan - seconds)
 
     def restrict(self, names: List[str]) -> "Plan":

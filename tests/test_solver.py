@@ -280,3 +280,20 @@ def test_convert_into_comprehensible_and_restrict(save_dir):
     sub = plan.restrict(["b"])
     assert sub.task_names == ["b"]
     assert len(sub.start_times) == 1 and len(sub.gpu_sets) == 1
+
+
+def test_shift_shrinks_running_task_runtime(save_dir):
+    """Fuzz-caught round 2: shifting a kept plan must consume the executed
+    portion of a running task's runtime, or its clamped start overlaps its
+    successors."""
+    a = make_task("sa", {1: 100.0}, save_dir)
+    b = make_task("sb", {1: 50.0}, save_dir)
+    plan = solve([a, b], n_gpus=1, timeout=10)
+    check_plan_valid(plan, [a, b], 1)
+    first = 0 if plan.start_times[0] < plan.start_times[1] else 1
+    plan.shift(30.0)
+    # the running (first) task lost 30 s of runtime, the queued one none
+    assert plan.runtimes[first] == pytest.approx(
+        (100.0 if plan.task_names[first] == "sa" else 50.0) - 30.0
+    )
+    check_plan_valid(plan, [a, b], 1)
