@@ -108,9 +108,16 @@ class _Unit:
         r = _rank()
         shard = flat[r * self.shard_numel : (r + 1) * self.shard_numel].clone()
         if offload:
-            shard = shard.to("cpu", copy=True)
-            if device.type == "cuda":
-                shard = shard.pin_memory()
+            # allocate the host shard pinned UP FRONT: .to("cpu") followed
+            # by .pin_memory() holds two full host copies at once — for a
+            # 70B model that is 2 x 140 GB of DRAM and can OOM the host
+            # (it killed a GPU box in round 2's config-5 run)
+            host = torch.empty(
+                shard.shape, dtype=shard.dtype, device="cpu",
+                pin_memory=device.type == "cuda",
+            )
+            host.copy_(shard)
+            shard = host
         # The optimizer updates this leaf directly (ZeRO: optimizer state is
         # sharded for free).
         self.shard = nn.Parameter(shard)
