@@ -30,7 +30,7 @@ def log(*a):
     print(*a, file=sys.stderr, flush=True)
 
 
-def build_tasks(config: int, scale: str, save_dir: str):
+def build_tasks(config: int, scale: str, save_dir: str, batches=None):
     from saturn_amd import HParams, Task
     from saturn_amd.models import (
         get_bert_model,
@@ -52,7 +52,7 @@ def build_tasks(config: int, scale: str, save_dir: str):
     )
 
     tiny = scale == "tiny"
-    bc = 8 if tiny else 64  # batches per job
+    bc = batches if batches is not None else (8 if tiny else 64)
 
     def t(name, gm, gd, loss, lr, gpu_range, hints=None, kwargs=None):
         return Task(
@@ -139,6 +139,8 @@ def main() -> None:
     ap.add_argument("--scale", choices=["tiny", "full"], default="tiny")
     ap.add_argument("--n-gpus", type=int, default=None)
     ap.add_argument("--interval", type=float, default=None)
+    ap.add_argument("--batches", type=int, default=None,
+                    help="batches per job (default 8 tiny / 64 full)")
     ap.add_argument(
         "--ckpt", choices=["all", "none"], default="all",
         help="'none' sets SATURN_SKIP_CKPT=1: measure makespan without "
@@ -175,7 +177,8 @@ def main() -> None:
         "pipeline": PipelineExecutor,
         "spilled": SpilledExecutor,
     }
-    tasks, execs = build_tasks(args.config, args.scale, save_dir)
+    tasks, execs = build_tasks(args.config, args.scale, save_dir,
+                               batches=args.batches)
     for nm in execs:
         register(nm, by_name[nm])
 
@@ -222,6 +225,7 @@ def main() -> None:
             "baseline_config": args.config,
             "scale": args.scale,
             "n_jobs": len(tasks),
+            "batches_per_job": tasks[0].total_batches,
             "techniques": execs,
         },
         "ckpt": args.ckpt,
