@@ -899,18 +899,21 @@ constexpr int QB = 64;  // q rows per block (4 waves x 16)
 constexpr int KT = 32;  // keys per tile iteration
 }  // namespace dq
 
-template <int D>
+template <int D, bool GKV>
 struct AttnDqLds {
-  unsigned short qt[dq::QB][D + 8];
-  unsigned short dot[dq::QB][D + 8];
+  // GKV (D=256): K/V/Q/dO fragments read straight from global (L2) and
+  // only the K tile is LDS-staged for the transposed dQ B-fragment —
+  // 97.7 KB -> 17.4 KB of LDS lifts occupancy from 1 to 3 blocks/CU.
+  unsigned short qt[GKV ? 1 : dq::QB][D + 8];
+  unsigned short dot[GKV ? 1 : dq::QB][D + 8];
   unsigned short kt[dq::KT][D + 8];
-  unsigned short vt[dq::KT][D + 8];
+  unsigned short vt[GKV ? 1 : dq::KT][D + 8];
   float lse_t[dq::QB];
   float delta_t[dq::QB];
 };
 
-template <int D, bool TR16>
-__launch_bounds__(256, D <= 128 ? 2 : 1)
+template <int D, bool TR16, bool GKV>
+__launch_bounds__(256, GKV ? 3 : (D <= 128 ? 2 : 1))
 __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
                                const unsigned short* __restrict__ Q,
                                const unsigned short* __restrict__ K,
@@ -923,7 +926,7 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
                                TStr vs, TStr dqs) {
   using namespace dq;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  AttnDqLds<D>& lds = *reinterpret_cast<AttnDqLds<D>*>(smem);
+  AttnDqLds<D, GKV>& lds = *reinterpret_cast<AttnDqLds<D, GKV>*>(smem);
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
@@ -940,18 +943,21 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
   const float* lse_h = LSE + bh * (long)T;
   const float* del_h = DELTA + bh * (long)T;
 
-  // ---- stage the block's Q / dO rows + lse/delta once
+  // ---- stage the block's Q / dO rows + lse/delta once (GKV reads the
+  // row fragments straight from L2 instead)
   {
-    constexpr int CHUNKS = (QB * D) / (256 * 8);
+    if constexpr (!GKV) {
+      constexpr int CHUNKS = (QB * D) / (256 * 8);
 #pragma unroll
-    for (int c = 0; c < CHUNKS; ++c) {
-      int idx = (c * 256 + threadIdx.x) * 8;
-      int row = idx / D, col = idx % D;
-      *reinterpret_cast<bf16x8*>(&lds.qt[row][col]) =
-          *reinterpret_cast<const bf16x8*>(Qh + (long)(q0 + row) * qs.st + col);
-      *reinterpret_cast<bf16x8*>(&lds.dot[row][col]) =
-          *reinterpret_cast<const bf16x8*>(
-              dOh + (long)(q0 + row) * dos.st + col);
+      for (int c = 0; c < CHUNKS; ++c) {
+        int idx = (c * 256 + threadIdx.x) * 8;
+        int row = idx / D, col = idx % D;
+        *reinterpret_cast<bf16x8*>(&lds.qt[row][col]) =
+            *reinterpret_cast<const bf16x8*>(Qh + (long)(q0 + row) * qs.st + col);
+        *reinterpret_cast<bf16x8*>(&lds.dot[row][col]) =
+            *reinterpret_cast<const bf16x8*>(
+                dOh + (long)(q0 + row) * dos.st + col);
+      }
     }
     if (threadIdx.x < QB) {
       lds.lse_t[threadIdx.x] = lse_h[q0 + threadIdx.x];
@@ -975,9 +981,10 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
         *reinterpret_cast<bf16x8*>(&lds.kt[row][col]) =
             *reinterpret_cast<const bf16x8*>(
                 Kh + (long)(k0 + row) * ks.st + col);
-        *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
-            *reinterpret_cast<const bf16x8*>(
-                Vh + (long)(k0 + row) * vs.st + col);
+        if constexpr (!GKV)
+          *reinterpret_cast<bf16x8*>(&lds.vt[row][col]) =
+              *reinterpret_cast<const bf16x8*>(
+                  Vh + (long)(k0 + row) * vs.st + col);
       }
     }
     __syncthreads();
@@ -996,12 +1003,22 @@ __global__ void attn_dq_kernel(const unsigned short* __restrict__ dO,
       for (int ds_ = 0; ds_ < D / 32; ++ds_) {
         bf16x8 ak = *reinterpret_cast<const bf16x8*>(
             &lds.kt[n * 16 + r][ds_ * 32 + qg * 8]);
-        bf16x8 av = *reinterpret_cast<const bf16x8*>(
-            &lds.vt[n * 16 + r][ds_ * 32 + qg * 8]);
-        bf16x8 bq = *reinterpret_cast<const bf16x8*>(
-            &lds.qt[wid * 16 + r][ds_ * 32 + qg * 8]);
-        bf16x8 bdo = *reinterpret_cast<const bf16x8*>(
-            &lds.dot[wid * 16 + r][ds_ * 32 + qg * 8]);
+        bf16x8 av, bq, bdo;
+        if constexpr (GKV) {
+          av = *reinterpret_cast<const bf16x8*>(
+              Vh + (long)(k0 + n * 16 + r) * vs.st + ds_ * 32 + qg * 8);
+          bq = *reinterpret_cast<const bf16x8*>(
+              Qh + (long)(q0 + wid * 16 + r) * qs.st + ds_ * 32 + qg * 8);
+          bdo = *reinterpret_cast<const bf16x8*>(
+              dOh + (long)(q0 + wid * 16 + r) * dos.st + ds_ * 32 + qg * 8);
+        } else {
+          av = *reinterpret_cast<const bf16x8*>(
+              &lds.vt[n * 16 + r][ds_ * 32 + qg * 8]);
+          bq = *reinterpret_cast<const bf16x8*>(
+              &lds.qt[wid * 16 + r][ds_ * 32 + qg * 8]);
+          bdo = *reinterpret_cast<const bf16x8*>(
+              &lds.dot[wid * 16 + r][ds_ * 32 + qg * 8]);
+        }
         st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak, bq, st, 0, 0, 0);
         dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bdo, dpt, 0, 0, 0);
       }
@@ -1118,6 +1135,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
     const char* e = getenv("SAMD_ATTN_BWD_TR16");
     return !(e && e[0] == '0');
   }();
+  static const bool use_dq_gkv = [] {
+    const char* e = getenv("SAMD_ATTN_DQ_GKV");
+    return e && e[0] == '1';
+  }();
   // V tile from L2 instead of LDS at D<=128 (SAMD_ATTN_BWD_VL2=1): trades
   // per-q-tile L2 re-reads for 62->45 KB LDS/block (3 blocks/CU scalar path)
   static const bool use_vl2 = [] {
@@ -1150,22 +1171,42 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                        reinterpret_cast<const unsigned short*>(o.data_ptr()), \
                        delta.data_ptr<float>(), T, H, str_of(dout),          \
                        str_of(o));                                           \
-    constexpr size_t SH_DQ = sizeof(AttnDqLds<DD>);                          \
+    /* GKV (K/V/Q/dO fragments from L2, 17 KB LDS, 3 blocks/CU) measured \
+       SLOWER at D=256 (612 vs 525 us at the GPT-J bench shape: the L2   \
+       re-reads cost more than the occupancy gains) — kept as an opt-in  \
+       probe only */                                                       \
+    const bool gkv = use_dq_gkv && DD > 128;                                 \
+    size_t SH_DQ = gkv ? sizeof(AttnDqLds<DD, true>)                         \
+                       : sizeof(AttnDqLds<DD, false>);                       \
     static bool dq_attr_##DD = [] {                                          \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&attn_dq_kernel<DD, false>),         \
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_DQ);           \
+          reinterpret_cast<const void*>(&attn_dq_kernel<DD, false, false>),  \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
+          (int)sizeof(AttnDqLds<DD, false>));                                \
       hipFuncSetAttribute(                                                   \
-          reinterpret_cast<const void*>(&attn_dq_kernel<DD, true>),          \
-          hipFuncAttributeMaxDynamicSharedMemorySize, (int)SH_DQ);           \
+          reinterpret_cast<const void*>(&attn_dq_kernel<DD, true, false>),   \
+          hipFuncAttributeMaxDynamicSharedMemorySize,                        \
+          (int)sizeof(AttnDqLds<DD, false>));                                \
       return true;                                                           \
     }();                                                                     \
     (void)dq_attr_##DD;                                                      \
     {                                                                        \
       dim3 qgrid(B * H, T / dq::QB), qblock(256);                            \
-      if (use_tr16 && DD <= 128)                                             \
-        hipLaunchKernelGGL((attn_dq_kernel<DD, true>), qgrid, qblock,        \
-                           SH_DQ, stream.stream(),                           \
+      if (gkv)                                                               \
+        hipLaunchKernelGGL((attn_dq_kernel<DD, false, true>), qgrid,         \
+                           qblock, SH_DQ, stream.stream(),                   \
+                           reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(q.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(k.data_ptr()), \
+                           reinterpret_cast<const unsigned short*>(v.data_ptr()), \
+                           lse.data_ptr<float>(), delta.data_ptr<float>(),   \
+                           reinterpret_cast<unsigned short*>(dq.data_ptr()), \
+                           T, H, Hkv, scale, causal ? 1 : 0, (int)kv_len,   \
+                           str_of(dout), str_of(q), str_of(k), str_of(v),    \
+                           str_of(dq));                                      \
+      else if (use_tr16 && DD <= 128)                                        \
+        hipLaunchKernelGGL((attn_dq_kernel<DD, true, false>), qgrid,         \
+                           qblock, SH_DQ, stream.stream(),                   \
                            reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
                            reinterpret_cast<const unsigned short*>(q.data_ptr()), \
                            reinterpret_cast<const unsigned short*>(k.data_ptr()), \
@@ -1176,8 +1217,8 @@ std::vector<at::Tensor> attn_bwd(at::Tensor dout, at::Tensor q, at::Tensor k,
                            str_of(dout), str_of(q), str_of(k), str_of(v),    \
                            str_of(dq));                                      \
       else                                                                   \
-        hipLaunchKernelGGL((attn_dq_kernel<DD, false>), qgrid, qblock,       \
-                           SH_DQ, stream.stream(),                           \
+        hipLaunchKernelGGL((attn_dq_kernel<DD, false, false>), qgrid,        \
+                           qblock, SH_DQ, stream.stream(),                   \
                            reinterpret_cast<const unsigned short*>(dout.data_ptr()), \
                            reinterpret_cast<const unsigned short*>(q.data_ptr()), \
                            reinterpret_cast<const unsigned short*>(k.data_ptr()), \
