@@ -78,7 +78,7 @@ __global__ void gelu_kernel(const T* __restrict__ x, const T* __restrict__ dy,
 static dim3 grid_1d(long work) {
   const int block = 256;
   return dim3((unsigned)std::max<long>(
-      1, std::min<long>((work + block - 1) / block, 4096)));
+      1, std::min<long>((work + block - 1) / block, 8192)));
 }
 
 at::Tensor gelu_fwd(at::Tensor x) {
