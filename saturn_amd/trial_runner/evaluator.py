@@ -223,9 +223,13 @@ def _profile_winners(tasks: List, n_gpus: int) -> None:
                 dill.dump((strat.executor, t, g, 10_000 + ti), fh)
             import sys as _sys
 
+            # Bounded: rocprofv3 tracing a cell that itself gang-spawns
+            # workers can fail to drain (observed on hardware) — a timeout
+            # degrades this task to timeit-only instead of wedging search
             rows = rocprof_stats(
                 [_sys.executable, "-m", "saturn_amd.trial_runner.cell_main",
-                 payload, result]
+                 payload, result],
+                timeout=240.0,
             )
         if rows:
             strat.parameters["kernels"] = [

@@ -79,9 +79,24 @@ def rocprof_stats(
         *cmd,
     ]
     try:
-        subprocess.run(
-            full, capture_output=True, text=True, timeout=timeout, cwd=workdir
+        # own process group: on timeout kill the WHOLE tree (rocprofv3's
+        # traced cell spawns gang workers; killing only the wrapper would
+        # leak them on the box)
+        proc = subprocess.Popen(
+            full,
+            stdout=subprocess.DEVNULL,
+            stderr=subprocess.DEVNULL,
+            cwd=workdir,
+            start_new_session=True,
         )
+        try:
+            proc.wait(timeout=timeout)
+        except subprocess.TimeoutExpired:
+            import signal
+
+            os.killpg(os.getpgid(proc.pid), signal.SIGKILL)
+            proc.wait(timeout=30)
+            return None
     except Exception:
         return None
     rows: List[Dict[str, str]] = []
