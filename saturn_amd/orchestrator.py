@@ -27,7 +27,7 @@ def orchestrate(
     interval: float = 1000.0,
     n_gpus: Optional[int] = None,
     solver_timeout: Optional[float] = None,
-    hysteresis: float = 500.0,
+    hysteresis: Optional[float] = None,
     launch_timeout: Optional[float] = None,
     max_task_retries: int = 2,
 ) -> None:
@@ -52,6 +52,11 @@ def orchestrate(
         n_gpus = detect_gpu_count()
     if solver_timeout is None:
         solver_timeout = max(1.0, interval / 2)
+    if hysteresis is None:
+        # the reference's 500 s constant is interval/2 of ITS default
+        # interval=1000 (milp.py:363,377); a fixed 500 makes introspection
+        # inert when intervals are seconds-scale
+        hysteresis = interval / 2
 
     task_list = list(task_list)
     infeasible = [
