@@ -109,9 +109,9 @@ class BucketedDDP(torch.nn.Module):
         if use_buckets is None:
             use_buckets = self.world > 1
         if not use_buckets:
-            # No comm -> no buckets.  Grad-as-bucket-view costs ~4 bytes/um
-            # of pure bookkeeping traffic per step (zero the flat + autograd
-            # accumulate-into-view instead of assign); at world 1 the flat
+            # No comm -> no buckets.  Grad-as-bucket-view costs ~4 bytes of
+            # pure bookkeeping HBM traffic per gradient element per step
+            # (zero the flat + autograd accumulate-into-view); at world 1 the flat
             # buffer serves nothing, so let autograd assign fresh grads and
             # zero with set_to_none.  (Measured: ~8 ms/step of FillFunctor +
             # CUDAFunctor_add on GPT-J-6B, profiles/r02_baseline_kernels.txt)
