@@ -184,3 +184,32 @@ def test_solve_empty_task_list():
 
     plan = solve([], n_gpus=4, timeout=5)
     assert plan.task_names == [] and plan.makespan == 0.0
+
+
+def test_skip_ckpt_env_and_delete(tmp_path, monkeypatch):
+    """SATURN_SKIP_CKPT=1 makes save_checkpoint a no-op (pure-makespan
+    benchmarking on small scratch disks); delete_checkpoint removes the
+    ckpt and any per-rank shard files."""
+    import torch
+
+    from saturn_amd import HParams, Task
+
+    t = Task(
+        lambda: torch.nn.Linear(2, 2),
+        lambda: [0],
+        lambda a, b: None,
+        HParams(lr=1e-3, batch_count=1),
+        name="ckpt_env",
+        save_dir=str(tmp_path),
+    )
+    m = torch.nn.Linear(2, 2)
+    monkeypatch.setenv("SATURN_SKIP_CKPT", "1")
+    t.save_checkpoint(m)
+    assert not t.has_ckpt()
+    monkeypatch.delenv("SATURN_SKIP_CKPT")
+    t.save_checkpoint(m)
+    assert t.has_ckpt()
+    shard = tmp_path / "ckpt_env.optshard.w2.r0.pt"
+    shard.write_bytes(b"x")
+    t.delete_checkpoint()
+    assert not t.has_ckpt() and not shard.exists()
