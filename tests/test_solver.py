@@ -297,3 +297,20 @@ def test_shift_shrinks_running_task_runtime(save_dir):
         (100.0 if plan.task_names[first] == "sa" else 50.0) - 30.0
     )
     check_plan_valid(plan, [a, b], 1)
+
+
+def test_timeout_keeps_restricted_incumbent(save_dir):
+    """When the warm-start bound proves no better plan exists and a task
+    retired, the shifted RESTRICTED previous plan is kept (not greedy)."""
+    a = make_task("ra", {1: 4000.0}, save_dir)
+    b = make_task("rb", {1: 3000.0}, save_dir)
+    old = solve([a, b], n_gpus=1, timeout=10)
+    check_plan_valid(old, [a, b], 1)
+    # b retires; re-solve over {a} with the restricted presolved plan and
+    # a runtime identical to before: nothing beats the incumbent
+    pre = old.restrict(["ra"])
+    plan = solve([a], presolved=pre, interval=500.0, n_gpus=1, timeout=10)
+    check_plan_valid(plan, [a], 1)
+    assert plan.solver_status in ("kept_incumbent", "optimal")
+    if plan.solver_status == "kept_incumbent":
+        assert plan.task_names == ["ra"]
