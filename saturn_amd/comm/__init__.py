@@ -45,6 +45,63 @@ def require_native_comm():
     return ext
 
 
+class _Comm:
+    """Thin keep-alive wrapper over the native RcclComm.
+
+    Collectives run on the engine's own HIP stream; a caller that drops a
+    tensor right after the call could otherwise have the caching allocator
+    reuse its memory while the collective still reads it (the hazard
+    torch's ProcessGroupNCCL solves with recordStream).  Holding a Python
+    reference until ``join()`` is the simple, allocator-agnostic fix.
+    """
+
+    def __init__(self, inner):
+        self._c = inner
+        self._pending = []
+
+    @property
+    def rank(self):
+        return self._c.rank
+
+    @property
+    def world(self):
+        return self._c.world
+
+    def all_reduce(self, t, average=True):
+        self._pending.append(t)
+        self._c.all_reduce(t, average)
+
+    def broadcast(self, t, root):
+        self._pending.append(t)
+        self._c.broadcast(t, root)
+
+    def all_gather(self, out, inp):
+        self._pending.extend((out, inp))
+        self._c.all_gather(out, inp)
+
+    def reduce_scatter(self, out, inp, average=False):
+        self._pending.extend((out, inp))
+        self._c.reduce_scatter(out, inp, average)
+
+    def send(self, t, peer):
+        self._pending.append(t)
+        self._c.send(t, peer)
+
+    def recv(self, t, peer):
+        self._pending.append(t)
+        self._c.recv(t, peer)
+
+    def join(self):
+        # compute stream now waits the comm stream: anything enqueued so
+        # far is ordered before future compute, so the references can drop
+        self._c.join()
+        self._pending.clear()
+
+    def synchronize(self):
+        self._c.synchronize()
+        self._pending.clear()
+
+
 def create_comm(rank: int, world: int):
     """Create a communicator for the current process group's gang.  Uses
     torch.distributed (already initialized by the executor launch) only to
@@ -53,8 +110,8 @@ def create_comm(rank: int, world: int):
     import torch.distributed as dist
 
     if world == 1:
-        return ext.RcclComm(ext.get_unique_id(), 0, 1)
+        return _Comm(ext.RcclComm(ext.get_unique_id(), 0, 1))
     assert dist.is_initialized(), "init_process_group first (id exchange)"
     obj = [ext.get_unique_id() if rank == 0 else None]
     dist.broadcast_object_list(obj, src=0)
-    return ext.RcclComm(obj[0], rank, world)
+    return _Comm(ext.RcclComm(obj[0], rank, world))

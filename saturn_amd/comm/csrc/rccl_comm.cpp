@@ -15,8 +15,6 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
-#include <ATen/hip/impl/HIPCachingAllocatorMasqueradingAsCUDA.h>
-#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
@@ -82,24 +80,9 @@ class RcclComm {
     HIP_CHECK(hipStreamWaitEvent(stream_, entry_ev_, 0));
   }
 
-  // The tensor's storage is consumed on the side stream: tell the caching
-  // allocator, or a caller that drops the tensor right after the call can
-  // have its memory reused while the collective still reads it (torch's
-  // ProcessGroupNCCL records streams the same way).
-  void record(const at::Tensor& t) {
-    // the Masquerading variants are what torch's allocator actually
-    // tracks on ROCm (the build hipifies this TU)
-    c10::hip::HIPCachingAllocatorMasqueradingAsCUDA::
-        recordStreamMasqueradingAsCUDA(
-            t.storage().data_ptr(),
-            c10::hip::getStreamFromExternalMasqueradingAsCUDA(
-                stream_, t.get_device()));
-  }
-
   void all_reduce(at::Tensor t, bool average) {
     TORCH_CHECK(t.is_cuda() && t.is_contiguous());
     fence_compute();
-    record(t);
     RCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
                              dtype_of(t), average ? ncclAvg : ncclSum, comm_,
                              stream_));
@@ -108,7 +91,6 @@ class RcclComm {
   void broadcast(at::Tensor t, int root) {
     TORCH_CHECK(t.is_cuda() && t.is_contiguous());
     fence_compute();
-    record(t);
     RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
                              dtype_of(t), root, comm_, stream_));
   }
@@ -117,8 +99,6 @@ class RcclComm {
     TORCH_CHECK(out.is_cuda() && in.is_cuda());
     TORCH_CHECK(out.numel() == in.numel() * world_);
     fence_compute();
-    record(in);
-    record(out);
     RCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), in.numel(),
                              dtype_of(in), comm_, stream_));
   }
@@ -127,8 +107,6 @@ class RcclComm {
     TORCH_CHECK(out.is_cuda() && in.is_cuda());
     TORCH_CHECK(in.numel() == out.numel() * world_);
     fence_compute();
-    record(in);
-    record(out);
     RCCL_CHECK(ncclReduceScatter(in.data_ptr(), out.data_ptr(), out.numel(),
                                  dtype_of(in), average ? ncclAvg : ncclSum,
                                  comm_, stream_));
@@ -136,14 +114,12 @@ class RcclComm {
 
   void send(at::Tensor t, int peer) {
     fence_compute();
-    record(t);
     RCCL_CHECK(ncclSend(t.data_ptr(), t.numel(), dtype_of(t), peer, comm_,
                         stream_));
   }
 
   void recv(at::Tensor t, int peer) {
     fence_compute();
-    record(t);
     RCCL_CHECK(ncclRecv(t.data_ptr(), t.numel(), dtype_of(t), peer, comm_,
                         stream_));
   }
