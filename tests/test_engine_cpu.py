@@ -235,3 +235,57 @@ def test_search_profile_attaches_kernel_evidence(save_dir, library_path, monkeyp
     assert "kernels" in best.parameters, best.parameters
     assert best.parameters["kernels"][0]["name"].startswith("samd::")
     assert "rocm_smi" in best.parameters
+
+
+def test_execute_serializes_shared_gpu_tasks(save_dir, tmp_path):
+    """Engine invariant: tasks sharing a GPU must run strictly in plan
+    order; disjoint-GPU tasks may overlap.  Verified with a file-logging
+    executor (the subprocess boundary is real)."""
+    import json
+    import time
+
+    from saturn_amd.core.technique import BaseTechnique
+    from saturn_amd.engine.gang import execute
+    from saturn_amd.solver.milp import Plan
+
+    logf = str(tmp_path / "exec_log.jsonl")
+
+    class Logger(BaseTechnique):
+        name = "logger"
+
+        @staticmethod
+        def execute(task, gpus, tid, batch_count):
+            t0 = time.monotonic()
+            time.sleep(0.4)
+            with open(logf, "a") as fh:
+                fh.write(json.dumps(
+                    {"task": task.name, "start": t0,
+                     "end": time.monotonic()}) + "\n")
+
+        @staticmethod
+        def search(task, gpus, tid):
+            return {"x": 1}, 0.1
+
+    tasks = [make_mlp_task(nm, save_dir) for nm in ("ea", "eb", "ec")]
+    for t in tasks:
+        t.strategies[1] = Strategy(Logger, 1, {"x": 1}, 1.0, batch_time=0.1)
+        t.select_strategy(t.strategies[1])
+    # ea then eb share GPU 0 (eb depends on ea); ec alone on GPU 1
+    plan = Plan(
+        task_names=["ea", "eb", "ec"],
+        chosen_option=[0, 0, 0],
+        gpu_counts=[1, 1, 1],
+        gpu_sets=[[0], [0], [1]],
+        start_times=[0.0, 1.0, 0.0],
+        runtimes=[1.0, 1.0, 1.0],
+        makespan=2.0,
+    )
+    execute(tasks, [1, 1, 1], 2.0, plan)
+    rows = {r["task"]: r for r in map(json.loads,
+                                      open(logf).read().splitlines())}
+    assert set(rows) == {"ea", "eb", "ec"}
+    # ea fully precedes eb (shared GPU)
+    assert rows["ea"]["end"] <= rows["eb"]["start"] + 1e-3
+    # ec overlaps ea (disjoint GPUs, both start at 0): it must NOT have
+    # waited for the whole ea+eb chain
+    assert rows["ec"]["start"] < rows["eb"]["start"]
