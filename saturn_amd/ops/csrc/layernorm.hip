@@ -171,8 +171,12 @@ __global__ void norm_bwd_vec_kernel(const T* __restrict__ dy,
   float dbp[8 * GMAX];
 #pragma unroll
   for (int k = 0; k < 8 * GMAX; ++k) { dwp[k] = 0.f; dbp[k] = 0.f; }
-  float c_dyw[8 * GMAX];
-  float c_xh[8 * GMAX];
+  // row cache only at GMAX<=2: at GMAX=4 the four arrays are 128 floats
+  // and the kernel spills ~200 B/lane to scratch (compiler-verified) —
+  // re-reading the row from HBM is cheaper than scratch traffic
+  constexpr bool CACHE = GMAX <= 2;
+  float c_dyw[CACHE ? 8 * GMAX : 1];
+  float c_xh[CACHE ? 8 * GMAX : 1];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* dyr = dy + (long)row * cols;
     const T* xr = x + (long)row * cols;
@@ -192,8 +196,10 @@ __global__ void norm_bwd_vec_kernel(const T* __restrict__ dy,
           const float dyf = elemF<T>(dyv, e);
           const float xh = (elemF<T>(xv, e) - mu) * rs;
           const float dyw = dyf * elemF<T>(wv, e);
-          c_dyw[g * 8 + e] = dyw;
-          c_xh[g * 8 + e] = xh;
+          if constexpr (CACHE) {
+            c_dyw[g * 8 + e] = dyw;
+            c_xh[g * 8 + e] = xh;
+          }
           s1 += dyw;
           s2 += dyw * xh;
           dwp[g * 8 + e] += dyf * xh;
@@ -208,10 +214,22 @@ __global__ void norm_bwd_vec_kernel(const T* __restrict__ dy,
       const int j = (g * LN_BLOCK + threadIdx.x) * 8;
       if (j < cols) {
         short8v_ln ov;
+        if constexpr (CACHE) {
 #pragma unroll
-        for (int e = 0; e < 8; ++e)
-          setElem<T>(ov, e,
-                     rs * (c_dyw[g * 8 + e] - m1 - c_xh[g * 8 + e] * m2));
+          for (int e = 0; e < 8; ++e)
+            setElem<T>(ov, e,
+                       rs * (c_dyw[g * 8 + e] - m1 - c_xh[g * 8 + e] * m2));
+        } else {
+          const short8v_ln dyv = *reinterpret_cast<const short8v_ln*>(dyr + j);
+          const short8v_ln xv = *reinterpret_cast<const short8v_ln*>(xr + j);
+          const short8v_ln wv = *reinterpret_cast<const short8v_ln*>(w + j);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const float dyw = elemF<T>(dyv, e) * elemF<T>(wv, e);
+            const float xh = (elemF<T>(xv, e) - mu) * rs;
+            setElem<T>(ov, e, rs * (dyw - m1 - xh * m2));
+          }
+        }
         *reinterpret_cast<short8v_ln*>(dxr + j) = ov;
       }
     }

@@ -148,7 +148,7 @@ def test_fused_adam_matches_adamw():
 # ---------------------------------------------------------------------------
 # LayerNorm / RMSNorm (K4)
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("cols", [1024, 4096, 16384])
+@pytest.mark.parametrize("cols", [1024, 4096, 8192, 16384])
 def test_layernorm_fwd_bwd(cols):
     requires_ext()
     from saturn_amd.ops.functional import fused_layer_norm
