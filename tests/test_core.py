@@ -213,3 +213,17 @@ def test_skip_ckpt_env_and_delete(tmp_path, monkeypatch):
     shard.write_bytes(b"x")
     t.delete_checkpoint()
     assert not t.has_ckpt() and not shard.exists()
+
+
+def test_port_pool_disjoint_and_env(monkeypatch):
+    """Per-task rendezvous ports must be disjoint for concurrent gangs and
+    always bind 127.0.0.1 (container hostnames may not resolve)."""
+    from saturn_amd.utils.ports import port_for, rendezvous_env
+
+    ports = {port_for(t) for t in range(64)}
+    assert len(ports) == 64  # no collisions across a realistic batch
+    env = rendezvous_env(7, rank=1, world_size=4)
+    assert env["MASTER_ADDR"] == "127.0.0.1"
+    assert env["RANK"] == "1" and env["WORLD_SIZE"] == "4"
+    monkeypatch.setenv("SATURN_PORT_BASE", "31000")
+    assert port_for(0) == 31000
