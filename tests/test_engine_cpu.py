@@ -289,3 +289,54 @@ def test_execute_serializes_shared_gpu_tasks(save_dir, tmp_path):
     # ec overlaps ea (disjoint GPUs, both start at 0): it must NOT have
     # waited for the whole ea+eb chain
     assert rows["ec"]["start"] < rows["eb"]["start"]
+
+
+def test_forecast_runtime_only_strategy(save_dir):
+    """A Strategy without batch_time falls back to runtime/total_batches
+    (cells imported from a plan dump may only carry whole-job runtime)."""
+    t = make_mlp_task("fr", save_dir, batch_count=10)
+    t.strategies[1] = Strategy(DDPExecutor, 1, {"bucket_mb": 1}, 20.0)
+    plan = solve([t], n_gpus=1, timeout=5)
+    t.select_strategy(t.strategies[1])
+    relevant, batches, _ = forecast([t], 10.0, plan)
+    # 10 s interval at 2 s/batch -> 5 batches
+    assert batches == [5]
+
+
+def test_execute_dependent_fails_when_dependency_fails(save_dir, tmp_path):
+    """A task whose shared-GPU predecessor crashed must fail (not run on a
+    half-trained checkpoint) and be reported for elastic retry."""
+    from saturn_amd.core.technique import BaseTechnique
+    from saturn_amd.engine.gang import execute
+    from saturn_amd.solver.milp import Plan
+
+    marker = str(tmp_path / "ran_b.txt")
+
+    class Boom(BaseTechnique):
+        name = "boom"
+
+        @staticmethod
+        def execute(task, gpus, tid, batch_count):
+            if task.name == "da":
+                raise RuntimeError("injected failure")
+            with open(marker, "w") as fh:
+                fh.write("ran")
+
+        @staticmethod
+        def search(task, gpus, tid):
+            return {"x": 1}, 0.1
+
+    tasks = [make_mlp_task(nm, save_dir) for nm in ("da", "db")]
+    for t in tasks:
+        t.strategies[1] = Strategy(Boom, 1, {"x": 1}, 1.0, batch_time=0.1)
+        t.select_strategy(t.strategies[1])
+    plan = Plan(
+        task_names=["da", "db"], chosen_option=[0, 0], gpu_counts=[1, 1],
+        gpu_sets=[[0], [0]], start_times=[0.0, 1.0], runtimes=[1.0, 1.0],
+        makespan=2.0,
+    )
+    failed = execute(tasks, [1, 1], 2.0, plan, raise_on_failure=False)
+    assert {t.name for t in failed} == {"da", "db"}
+    import os as _os
+
+    assert not _os.path.exists(marker)  # db never ran
