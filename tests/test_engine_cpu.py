@@ -340,3 +340,37 @@ def test_execute_dependent_fails_when_dependency_fails(save_dir, tmp_path):
     import os as _os
 
     assert not _os.path.exists(marker)  # db never ran
+
+
+def test_gpu_pool_never_oversubscribes():
+    """Trial-runner GPU-slot allocator: concurrent cells must never hold
+    overlapping device ids and blocked acquires proceed after release."""
+    import threading
+    import time
+
+    from saturn_amd.trial_runner.evaluator import _GpuPool
+
+    pool = _GpuPool(4)
+    held, errors = [], []
+    lock = threading.Lock()
+
+    def worker(g):
+        ids = pool.acquire(g)
+        with lock:
+            for other in held:
+                if set(ids) & set(other):
+                    errors.append((ids, other))
+            held.append(ids)
+        time.sleep(0.05)
+        with lock:
+            held.remove(ids)
+        pool.release(ids)
+
+    threads = [threading.Thread(target=worker, args=(g,))
+               for g in (2, 2, 1, 3, 4, 1, 2)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors, errors
+    assert pool.free == set(range(4))
