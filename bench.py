@@ -101,8 +101,12 @@ def main() -> None:
 
     torch.manual_seed(1234)
     log(f"[rank {rank}] building {args.model} on {device}...")
-    # build directly on the device: at world 8 a host-side build would
-    # transiently hold 8 x 24 GB fp32 replicas in DRAM
+    # build directly on the device IN the compute dtype: at world 8 a
+    # host-side build would transiently hold 8 x 24 GB fp32 replicas in
+    # DRAM, and fp32-on-device doubles the init writes
+    prev_dtype = torch.get_default_dtype()
+    if use_gpu:
+        torch.set_default_dtype(dtype)
     with device:
         if args.model == "llama-3-8b":
             from saturn_amd.models.llama import (
@@ -131,6 +135,7 @@ def main() -> None:
             )
             model = GPTJForCausalLM(mcfg)
             vocab = mcfg.vocab_size
+    torch.set_default_dtype(prev_dtype)
     model = model.to(dtype=dtype)
     model.train()
     n_params = sum(p.numel() for p in model.parameters())
