@@ -227,3 +227,22 @@ def test_port_pool_disjoint_and_env(monkeypatch):
     assert env["RANK"] == "1" and env["WORLD_SIZE"] == "4"
     monkeypatch.setenv("SATURN_PORT_BASE", "31000")
     assert port_for(0) == 31000
+
+
+def test_bench_makespan_configs_construct(tmp_path):
+    """All five BASELINE config task batches must construct on CPU without
+    instantiating models (Task holds factories only)."""
+    import os
+    import sys
+
+    sys.path.insert(
+        0, os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    )
+    from bench_makespan import build_tasks
+
+    for cfg in (1, 2, 3, 4, 5):
+        tasks, execs = build_tasks(cfg, "tiny", str(tmp_path), batches=2)
+        assert tasks and execs, cfg
+        for t in tasks:
+            assert callable(t.internal_get_model)
+            assert t.total_batches == 2
